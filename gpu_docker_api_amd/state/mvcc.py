@@ -1,0 +1,266 @@
+"""In-process MVCC key-value store with etcd v3 revision semantics.
+
+This is the contract everything else leans on: the
+reference's history/rollback features walk etcd per-key revisions
+(/root/reference/internal/etcd/revision.go:18-66), so the semantics here must
+match etcd exactly:
+
+* A single store-wide **revision** counter starts at 1; every mutating
+  transaction writes at ``current_revision + 1`` and advances the counter
+  (so the first put lands at revision 2, as in etcd).
+* Each key-value carries ``create_revision`` (revision at which the key was
+  created in its current lifetime), ``mod_revision`` (revision of the last
+  write) and ``version`` (number of writes since creation; resets when a key
+  is deleted and re-created).
+* ``get(key, rev=r)`` returns the key-value **as of revision r** — the state
+  after all transactions with revision <= r — exactly what etcd's
+  ``clientv3.WithRev`` does and what the reference's revision walker relies on.
+* ``compact(rev)`` discards history below ``rev``; reads below the compaction
+  point raise :class:`RevisionCompacted` (etcd: ErrCompacted).
+
+Thread-safe (one ``RLock``); an optional write-ahead log (``wal.py``) makes it
+durable so a single-node deployment needs no external etcd.
+"""
+from __future__ import annotations
+
+import threading
+from dataclasses import dataclass
+from typing import Callable, Dict, List, Optional, Tuple
+
+from ..xerrors import NotExistInStore, RevisionCompacted
+
+
+@dataclass(frozen=True)
+class KeyValue:
+    """One key-value as etcd would return it."""
+
+    key: str
+    value: str
+    create_revision: int
+    mod_revision: int
+    version: int
+
+    def to_dict(self) -> dict:
+        return {
+            "key": self.key,
+            "value": self.value,
+            "create_revision": self.create_revision,
+            "mod_revision": self.mod_revision,
+            "version": self.version,
+        }
+
+
+@dataclass(frozen=True)
+class _Event:
+    """One write to a key: a put (value is str) or a tombstone (value None)."""
+
+    mod_revision: int
+    value: Optional[str]
+    version: int          # etcd per-key Version at this event (0 for tombstone)
+    create_revision: int  # lifetime start (0 for tombstone)
+
+
+class MemoryMVCC:
+    """The store. Mutations may be observed via ``on_event`` (used by the WAL)."""
+
+    def __init__(self) -> None:
+        self._lock = threading.RLock()
+        self._rev = 1            # etcd: store starts at revision 1
+        self._compacted = 0      # highest compacted revision (exclusive floor)
+        self._hist: Dict[str, List[_Event]] = {}
+        self.on_event: Optional[Callable[[str, int, Optional[str]], None]] = None
+
+    # ------------------------------------------------------------------ info
+    @property
+    def revision(self) -> int:
+        with self._lock:
+            return self._rev
+
+    @property
+    def compacted_revision(self) -> int:
+        with self._lock:
+            return self._compacted
+
+    # ----------------------------------------------------------------- write
+    def put(self, key: str, value: str) -> KeyValue:
+        with self._lock:
+            rev = self._rev + 1
+            events = self._hist.setdefault(key, [])
+            last = events[-1] if events else None
+            if last is None or last.value is None:
+                create_rev, version = rev, 1
+            else:
+                create_rev, version = last.create_revision, last.version + 1
+            events.append(_Event(rev, value, version, create_rev))
+            self._rev = rev
+            if self.on_event:
+                self.on_event(key, rev, value)
+            return KeyValue(key, value, create_rev, rev, version)
+
+    def delete(self, key: str) -> int:
+        """Delete a key. Returns the number of keys deleted (0 or 1).
+
+        A successful delete advances the revision and leaves a tombstone, so
+        ``get(key, rev=old)`` still sees pre-delete values (etcd behavior).
+        """
+        with self._lock:
+            events = self._hist.get(key)
+            if not events or events[-1].value is None:
+                return 0
+            rev = self._rev + 1
+            events.append(_Event(rev, None, 0, 0))
+            self._rev = rev
+            if self.on_event:
+                self.on_event(key, rev, None)
+            return 1
+
+    def delete_prefix(self, prefix: str) -> int:
+        """Delete all live keys under a prefix in ONE transaction (etcd
+        DeleteRange semantics: a single revision bump for the whole range)."""
+        with self._lock:
+            live = [
+                k
+                for k, ev in self._hist.items()
+                if k.startswith(prefix) and ev and ev[-1].value is not None
+            ]
+            if not live:
+                return 0
+            rev = self._rev + 1
+            for k in sorted(live):
+                self._hist[k].append(_Event(rev, None, 0, 0))
+                if self.on_event:
+                    self.on_event(k, rev, None)
+            self._rev = rev
+            return len(live)
+
+    # ------------------------------------------------------------------ read
+    def get(self, key: str, rev: int = 0) -> KeyValue:
+        """Latest value (rev=0) or value as of revision ``rev``.
+
+        Raises NotExistInStore if absent (at that revision), RevisionCompacted
+        if ``rev`` is at or below the compaction point.
+        """
+        with self._lock:
+            kv = self._get_locked(key, rev)
+            if kv is None:
+                raise NotExistInStore(key)
+            return kv
+
+    def get_or_none(self, key: str, rev: int = 0) -> Optional[KeyValue]:
+        with self._lock:
+            return self._get_locked(key, rev)
+
+    def _get_locked(self, key: str, rev: int) -> Optional[KeyValue]:
+        if rev:
+            if rev <= self._compacted:
+                raise RevisionCompacted(f"revision {rev} compacted at {self._compacted}")
+            if rev > self._rev:
+                # etcd returns ErrFutureRev; surface as missing-with-context
+                raise NotExistInStore(f"{key}@{rev} (future revision, store at {self._rev})")
+        events = self._hist.get(key)
+        if not events:
+            return None
+        if rev == 0:
+            ev = events[-1]
+        else:
+            ev = None
+            for e in reversed(events):
+                if e.mod_revision <= rev:
+                    ev = e
+                    break
+            if ev is None:
+                return None
+        if ev.value is None:
+            return None
+        return KeyValue(key, ev.value, ev.create_revision, ev.mod_revision, ev.version)
+
+    def range_prefix(self, prefix: str, rev: int = 0) -> List[KeyValue]:
+        """All live key-values under a prefix (as of ``rev`` if nonzero),
+        sorted by key (etcd Range default order)."""
+        with self._lock:
+            out: List[KeyValue] = []
+            for key in sorted(self._hist):
+                if not key.startswith(prefix):
+                    continue
+                kv = self._get_locked(key, rev)
+                if kv is not None:
+                    out.append(kv)
+            return out
+
+    # -------------------------------------------------------------- history
+    def history(self, key: str) -> List[KeyValue]:
+        """All surviving put-revisions of a key, newest first.
+
+        Matches what the reference's walker extracts: it walks rev from
+        mod_revision down to create_revision issuing one Get(WithRev) per
+        revision and dedupes by per-key Version (etcd/revision.go:18-44).
+        We return the same set directly — every put event of the key's
+        current lifetime that is above the compaction point. O(#events),
+        not O(#store revisions): this is the "hot loop" the reference made
+        O(total revisions) round-trips for.
+        """
+        with self._lock:
+            events = self._hist.get(key)
+            if not events or events[-1].value is None:
+                raise NotExistInStore(key)
+            create_rev = events[-1].create_revision
+            out = [
+                KeyValue(key, e.value, e.create_revision, e.mod_revision, e.version)
+                for e in reversed(events)
+                if e.value is not None
+                and e.create_revision == create_rev
+                and e.mod_revision > self._compacted
+            ]
+            return out
+
+    def get_version(self, key: str, version: int) -> KeyValue:
+        """The key-value whose per-key Version equals ``version`` (current
+        lifetime) — the reference's GetRevision(version) walk
+        (etcd/revision.go:46-66) without the O(revisions) round trips."""
+        for kv in self.history(key):
+            if kv.version == version:
+                return kv
+        raise NotExistInStore(f"{key} version={version}")
+
+    # ------------------------------------------------------------ compaction
+    def compact(self, rev: int) -> None:
+        """Discard event history with mod_revision < rev (etcd keeps the
+        latest event at/below the compaction point per key)."""
+        with self._lock:
+            if rev > self._rev:
+                raise NotExistInStore(f"cannot compact future revision {rev}")
+            if rev <= self._compacted:
+                raise RevisionCompacted(f"already compacted at {self._compacted}")
+            for key, events in list(self._hist.items()):
+                keep: List[_Event] = []
+                # keep the newest event <= rev (it defines state at rev) ...
+                floor = None
+                for e in events:
+                    if e.mod_revision <= rev:
+                        floor = e
+                    else:
+                        keep.append(e)
+                if floor is not None and (floor.value is not None or keep):
+                    keep.insert(0, floor)
+                if keep:
+                    self._hist[key] = keep
+                else:
+                    del self._hist[key]
+            self._compacted = rev - 1 if rev > 0 else 0
+
+    # ---------------------------------------------------------------- replay
+    def replay(self, key: str, rev: int, value: Optional[str]) -> None:
+        """Re-apply a WAL record. Only valid in ascending rev order."""
+        with self._lock:
+            events = self._hist.setdefault(key, [])
+            last = events[-1] if events else None
+            if value is None:
+                events.append(_Event(rev, None, 0, 0))
+            else:
+                if last is None or last.value is None:
+                    create_rev, version = rev, 1
+                else:
+                    create_rev, version = last.create_revision, last.version + 1
+                events.append(_Event(rev, value, version, create_rev))
+            if rev > self._rev:
+                self._rev = rev
