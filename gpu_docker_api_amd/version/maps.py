@@ -1,0 +1,116 @@
+"""Version bookkeeping: name -> latest version, versioned-name -> merge path.
+
+Parity with the reference (internal/version/version.go:59-79, merge.go:41-57)
+with its race fixed: the reference mutates plain Go maps from concurrent HTTP
+handlers with no mutex (version.go:26 — SURVEY.md §5.2); here every map is
+lock-guarded. Persisted JSON shapes are identical (a flat name->int64 map
+under ``versions/{container,volume}VersionMapKey``; a flat versionedName->path
+map under ``merges/containerMergeMapKey``), persisted write-behind on every
+mutation and synchronously at shutdown.
+"""
+from __future__ import annotations
+
+import json
+import threading
+from typing import Dict, Optional
+
+from ..state.keys import Resource
+from ..state.store import StateStore
+from ..state.workqueue import WorkQueue
+
+
+class _PersistedMap:
+    resource: Resource
+
+    def __init__(self, store: StateStore, queue: Optional[WorkQueue], state_key: str) -> None:
+        self._store = store
+        self._queue = queue
+        self._key = state_key
+        self._lock = threading.RLock()
+        self._m: Dict[str, object] = {}
+
+    async def load(self) -> None:
+        kv = await self._store.get_or_none(self.resource, self._key)
+        if kv is not None:
+            self._m = dict(json.loads(kv.value))
+
+    def serialize(self) -> str:
+        with self._lock:
+            return json.dumps(self._m, separators=(",", ":"))
+
+    def _persist_async(self) -> None:
+        if self._queue is not None:
+            self._queue.put(self.resource, self._key, self.serialize())
+
+    async def persist(self) -> None:
+        await self._store.put(self.resource, self._key, self.serialize())
+
+
+class VersionMap(_PersistedMap):
+    """name -> latest version number (containers or volumes)."""
+
+    resource = Resource.VERSIONS
+
+    def get(self, name: str) -> Optional[int]:
+        with self._lock:
+            v = self._m.get(name)
+            return int(v) if v is not None else None
+
+    def set(self, name: str, version: int) -> None:
+        with self._lock:
+            self._m[name] = int(version)
+            self._persist_async()
+
+    def bump(self, name: str) -> int:
+        """Atomically increment-and-get (the reference does read+set in two
+        steps from unsynchronized handlers, replicaset_nomock.go:27-29)."""
+        with self._lock:
+            v = int(self._m.get(name, 0)) + 1
+            self._m[name] = v
+            self._persist_async()
+            return v
+
+    def remove(self, name: str) -> None:
+        with self._lock:
+            self._m.pop(name, None)
+            self._persist_async()
+
+    def exists(self, name: str) -> bool:
+        with self._lock:
+            return name in self._m
+
+    def snapshot(self) -> Dict[str, int]:
+        with self._lock:
+            return {k: int(v) for k, v in self._m.items()}
+
+
+class MergeMap(_PersistedMap):
+    """versioned container name -> preserved rootfs (merged-layer) path."""
+
+    resource = Resource.MERGES
+
+    def get(self, versioned_name: str) -> Optional[str]:
+        with self._lock:
+            v = self._m.get(versioned_name)
+            return str(v) if v is not None else None
+
+    def set(self, versioned_name: str, path: str) -> None:
+        with self._lock:
+            self._m[versioned_name] = path
+            self._persist_async()
+
+    def remove(self, versioned_name: str) -> None:
+        with self._lock:
+            self._m.pop(versioned_name, None)
+            self._persist_async()
+
+    def remove_prefix(self, name_dash: str) -> None:
+        """Drop every version entry of one replicaSet (delete flow)."""
+        with self._lock:
+            for k in [k for k in self._m if k.startswith(name_dash)]:
+                self._m.pop(k, None)
+            self._persist_async()
+
+    def snapshot(self) -> Dict[str, str]:
+        with self._lock:
+            return {k: str(v) for k, v in self._m.items()}
