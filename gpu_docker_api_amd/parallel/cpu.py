@@ -74,6 +74,20 @@ class CpuScheduler(BaseScheduler):
             self._persist_async()
             return ",".join(str(c) for c in chosen)
 
+    def apply_specific(self, cpuset: str | List[str]) -> None:
+        """Re-acquire an exact cpuset (startup of a stopped container)."""
+        ids = cpuset.split(",") if isinstance(cpuset, str) else list(cpuset)
+        ids = [i.strip() for i in ids if i and i.strip()]
+        if not ids:
+            return
+        with self._lock:
+            busy = [i for i in ids if self.cpu_status_map.get(i, 1) == 1]
+            if busy:
+                raise CpuNotEnough(f"CPUs already allocated: {busy}")
+            for i in ids:
+                self.cpu_status_map[i] = 1
+            self._persist_async()
+
     def restore(self, cpuset: str | List[str]) -> None:
         ids = cpuset.split(",") if isinstance(cpuset, str) else list(cpuset)
         ids = [i.strip() for i in ids if i and i.strip()]

@@ -104,6 +104,24 @@ class GpuScheduler(BaseScheduler):
             self._persist_async()
             return uuids
 
+    def apply_specific(self, uuids: List[str]) -> None:
+        """Re-acquire an exact GPU set (startup of a stopped container,
+        which released its resources — reference just restarts without
+        re-acquiring, letting a stopped container's GPUs be double-booked).
+        Raises GpuNotEnough if any is already allocated."""
+        if not uuids:
+            return
+        with self._lock:
+            busy = [u for u in uuids if self.gpu_status_map.get(u, 1) == 1]
+            if busy:
+                raise GpuNotEnough(f"GPUs already allocated: {busy}")
+            unknown = [u for u in uuids if u not in self.gpu_status_map]
+            if unknown:
+                raise GpuNotEnough(f"unknown GPUs: {unknown}")
+            for u in uuids:
+                self.gpu_status_map[u] = 1
+            self._persist_async()
+
     def restore(self, uuids: List[str]) -> None:
         if not uuids:
             return

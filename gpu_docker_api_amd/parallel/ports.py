@@ -101,6 +101,19 @@ class PortScheduler(BaseScheduler):
             self._persist_async()
             return chosen
 
+    def apply_specific(self, ports: List[int] | List[str]) -> None:
+        """Re-acquire exact host ports (startup of a stopped container)."""
+        ints = [int(p) for p in ports]
+        if not ints:
+            return
+        with self._lock:
+            busy = [p for p in ints if p in self.used]
+            if busy:
+                raise PortNotEnough(f"ports already allocated: {busy}")
+            for p in ints:
+                self.used.add(p)
+            self._persist_async()
+
     def restore(self, ports: List[int] | List[str]) -> None:
         if not ports:
             return

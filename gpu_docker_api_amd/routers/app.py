@@ -1,0 +1,198 @@
+"""Daemon wiring (the analog of the reference's program.Init/Start/Stop,
+/root/reference/cmd/gpu-docker-api/main.go:53-154).
+
+Init order mirrors main.go: state store -> work queue -> GPU/CPU/port
+schedulers -> version maps -> merges dir -> services -> HTTP routes.
+Shutdown drains the work queue then persists all scheduler/version state
+synchronously (main.go:139-154) — plus closes the runtime and the WAL.
+"""
+from __future__ import annotations
+
+import logging
+import os
+from contextlib import asynccontextmanager
+from typing import Optional
+
+from fastapi import FastAPI
+from fastapi.responses import PlainTextResponse
+
+from ..config import Config
+from ..parallel import CpuScheduler, GpuScheduler, PortScheduler, make_inventory
+from ..parallel.topology import Topology
+from ..runtime import make_runtime
+from ..services import ReplicaSetService, VolumeService
+from ..state.etcd_gateway import EtcdGatewayStore
+from ..state.keys import (
+    CONTAINER_MERGE_MAP_KEY,
+    CONTAINER_VERSION_MAP_KEY,
+    VOLUME_VERSION_MAP_KEY,
+)
+from ..state.store import MemoryStore, StateStore
+from ..state.workqueue import WorkQueue
+from ..utils.copy import CopyEngine
+from ..utils.timing import METRICS
+from ..version import MergeMap, VersionMap
+from . import replicaset as replicaset_router
+from . import resource as resource_router
+from . import volume as volume_router
+from .middleware import AuthMiddleware, CorsMiddleware
+
+log = logging.getLogger(__name__)
+
+
+class Daemon:
+    """Holds every initialized subsystem; built once per process."""
+
+    def __init__(self, cfg: Config) -> None:
+        self.cfg = cfg
+        self.store: Optional[StateStore] = None
+        self.queue: Optional[WorkQueue] = None
+        self.gpu: Optional[GpuScheduler] = None
+        self.cpu: Optional[CpuScheduler] = None
+        self.ports: Optional[PortScheduler] = None
+        self.container_versions: Optional[VersionMap] = None
+        self.volume_versions: Optional[VersionMap] = None
+        self.merges: Optional[MergeMap] = None
+        self.runtime = None
+        self.replicaset: Optional[ReplicaSetService] = None
+        self.volume: Optional[VolumeService] = None
+
+    async def start(self) -> None:
+        cfg = self.cfg
+        os.makedirs(cfg.data_dir, exist_ok=True)
+        os.makedirs(cfg.merges_dir, exist_ok=True)
+
+        if cfg.state.startswith("etcd:"):
+            self.store = EtcdGatewayStore(cfg.state.split(":", 1)[1] or "http://127.0.0.1:2379")
+        elif cfg.state == "memory":
+            self.store = MemoryStore(wal_path=cfg.wal_path)
+        else:
+            self.store = MemoryStore()
+        self.queue = WorkQueue(self.store)
+        self.queue.start()
+
+        inventory = make_inventory(cfg.inventory, cfg.mock_gpus)
+        probe = None
+        if cfg.probe_cache:
+            probe = Topology.load_probe_file(cfg.probe_cache)
+        if probe is None and cfg.run_xgmi_probe:
+            probe = await self._run_probe()
+        self.gpu = await GpuScheduler.create(self.store, self.queue, inventory, probe=probe)
+        self.cpu = await CpuScheduler.create(self.store, self.queue)
+        self.ports = await PortScheduler.create(
+            self.store, self.queue, cfg.port_start, cfg.port_end
+        )
+
+        self.container_versions = VersionMap(self.store, self.queue, CONTAINER_VERSION_MAP_KEY)
+        self.volume_versions = VersionMap(self.store, self.queue, VOLUME_VERSION_MAP_KEY)
+        self.merges = MergeMap(self.store, self.queue, CONTAINER_MERGE_MAP_KEY)
+        await self.container_versions.load()
+        await self.volume_versions.load()
+        await self.merges.load()
+
+        rt_kwargs = {}
+        if cfg.runtime in ("proc",):
+            rt_kwargs = {
+                "base_dir": os.path.join(cfg.data_dir, "procrt"),
+                "gpu_resolver": self.gpu.info_by_uuid,
+            }
+        elif cfg.runtime == "mock":
+            rt_kwargs = {"base_dir": os.path.join(cfg.data_dir, "mockrt")}
+        elif cfg.runtime == "docker":
+            rt_kwargs = {
+                "socket_path": cfg.docker_socket,
+                "gpu_resolver": self.gpu.info_by_uuid,
+            }
+        self.runtime = make_runtime(cfg.runtime, **rt_kwargs)
+
+        copy_engine = CopyEngine(cfg.copy_engine)
+        self.replicaset = ReplicaSetService(
+            store=self.store,
+            queue=self.queue,
+            gpu=self.gpu,
+            cpu=self.cpu,
+            ports=self.ports,
+            versions=self.container_versions,
+            merges=self.merges,
+            runtime=self.runtime,
+            copy_engine=copy_engine,
+            cfg=cfg,
+        )
+        self.volume = VolumeService(
+            store=self.store,
+            queue=self.queue,
+            versions=self.volume_versions,
+            runtime=self.runtime,
+            copy_engine=copy_engine,
+            cfg=cfg,
+        )
+        if cfg.run_rccl_smoke:
+            await self._run_rccl_smoke()
+
+    async def _run_probe(self) -> Optional[dict]:
+        """Run the native xGMI bandwidth probe (csrc -> ops.hipcore)."""
+        try:
+            from ..ops import hipcore
+
+            return await hipcore.run_probe_async()
+        except Exception as exc:  # no GPU / extension not built
+            log.warning("xGMI probe unavailable: %s", exc)
+            return None
+
+    async def _run_rccl_smoke(self) -> None:
+        try:
+            from ..ops import hipcore
+
+            result = await hipcore.run_rccl_smoke_async()
+            log.info("RCCL smoke: %s", result)
+        except Exception as exc:
+            log.warning("RCCL smoke unavailable: %s", exc)
+
+    async def stop(self) -> None:
+        # drain async writes, then persist everything synchronously
+        if self.queue is not None:
+            await self.queue.close()
+        for part in (self.gpu, self.cpu, self.ports):
+            if part is not None:
+                await part.persist()
+        for part in (self.container_versions, self.volume_versions, self.merges):
+            if part is not None:
+                await part.persist()
+        if self.runtime is not None:
+            await self.runtime.close()
+        if self.store is not None:
+            await self.store.close()
+
+
+def build_app(cfg: Optional[Config] = None, daemon: Optional[Daemon] = None) -> FastAPI:
+    cfg = cfg or Config()
+    d = daemon or Daemon(cfg)
+
+    @asynccontextmanager
+    async def lifespan(app: FastAPI):
+        if d.replicaset is None:
+            await d.start()
+        app.state.daemon = d
+        _mount(app, d)
+        yield
+        await d.stop()
+
+    app = FastAPI(title="gpu-docker-api-amd", lifespan=lifespan)
+    app.add_middleware(CorsMiddleware)
+    app.add_middleware(AuthMiddleware, apikey=cfg.apikey)
+
+    @app.get("/ping")
+    async def ping():
+        return {"message": "pong"}
+
+    @app.get("/metrics")
+    async def metrics():
+        return PlainTextResponse(METRICS.prometheus_text())
+
+    return app
+
+
+def _mount(app: FastAPI, d: Daemon) -> None:
+    app.include_router(replicaset_router.make_router(d.replicaset))
+    app.include_router(volume_router.make_router(d.volume))
+    app.include_router(resource_router.make_router(d.gpu, d.cpu, d.ports))

@@ -1,0 +1,95 @@
+"""Volume routes — 5 endpoints under /api/v1/volumes
+(reference: internal/routers/volume.go:19-25; validation :37-47, :84-90)."""
+from __future__ import annotations
+
+from fastapi import APIRouter, Request
+from pydantic import ValidationError
+
+from ..models import VolumeCreate, VolumeSize
+from ..models.memory import parse_size
+from ..services.volume import VolumeService
+from .codes import Code
+from .errors import log_error, map_error
+from .response import error, success
+
+
+def make_router(svc: VolumeService) -> APIRouter:
+    r = APIRouter(prefix="/api/v1/volumes")
+
+    async def _parse(request: Request, model):
+        try:
+            return model.model_validate(await request.json())
+        except (ValidationError, ValueError):
+            return None
+
+    def _valid_size(s: str) -> bool:
+        try:
+            parse_size(s)
+            return True
+        except Exception:
+            return False
+
+    @r.post("")
+    async def create(request: Request):
+        req = await _parse(request, VolumeCreate)
+        if req is None:
+            return error(Code.INVALID_PARAMS)
+        if not req.name:
+            return error(Code.VOLUME_NAME_EMPTY)
+        if "-" in req.name:
+            return error(Code.VOLUME_NAME_DASH)
+        if req.name.startswith("/"):
+            return error(Code.VOLUME_NAME_SLASH)
+        if req.size and not _valid_size(req.size):
+            return error(Code.VOLUME_SIZE_UNIT)
+        try:
+            data = await svc.create_volume(req)
+        except Exception as exc:  # noqa: BLE001
+            log_error("volume.create", exc)
+            return error(map_error(exc, Code.VOLUME_CREATE_FAILED, volume=True))
+        return success(data)
+
+    @r.patch("/{name}/size")
+    async def patch_size(name: str, request: Request):
+        req = await _parse(request, VolumeSize)
+        if req is None:
+            return error(Code.INVALID_PARAMS)
+        if not req.size or not _valid_size(req.size):
+            return error(Code.VOLUME_SIZE_UNIT)
+        try:
+            data = await svc.patch_volume_size(name, req.size)
+        except Exception as exc:  # noqa: BLE001
+            log_error("volume.patch", exc)
+            return error(map_error(exc, Code.VOLUME_PATCH_FAILED, volume=True))
+        return success(data)
+
+    @r.delete("/{name}")
+    async def delete(name: str, request: Request):
+        # ?noall present => keep the store record (reference volume.go:123-129)
+        keep = "noall" in request.query_params
+        try:
+            await svc.delete_volume(name, keep_record=keep)
+        except Exception as exc:  # noqa: BLE001
+            log_error("volume.delete", exc)
+            return error(map_error(exc, Code.VOLUME_DELETE_FAILED, volume=True))
+        return success(None)
+
+    @r.get("/{name}")
+    async def info(name: str):
+        try:
+            data = await svc.get_volume_info(name)
+        except Exception as exc:  # noqa: BLE001
+            log_error("volume.info", exc)
+            return error(map_error(exc, Code.VOLUME_GET_INFO_FAILED, volume=True))
+        return success(data)
+
+    @r.get("/{name}/history")
+    async def history(name: str):
+        try:
+            data = await svc.get_volume_history(name)
+        except Exception as exc:  # noqa: BLE001
+            log_error("volume.history", exc)
+            return error(map_error(exc, Code.VOLUME_GET_HISTORY_FAILED, volume=True))
+        return success(data)
+
+    return r

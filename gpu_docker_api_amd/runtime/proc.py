@@ -1,0 +1,326 @@
+"""Native process runtime ("proc").
+
+A lightweight container runtime with no dockerd dependency: each container is
+a supervised host process group with
+
+* a private rootfs directory (materialized from the image store) that plays
+  the role of the overlay2 writable layer — migration/commit operate on it
+  exactly like the docker driver's UpperDir
+  (/root/reference/utils/copy.go:48-54);
+* GPU isolation via ``ROCR_VISIBLE_DEVICES``/``HIP_VISIBLE_DEVICES`` computed
+  from the allocated GPU UUIDs — the ROCm-native equivalent of device
+  injection, enforced by the ROCm runtime itself;
+* best-effort cgroup-v2 limits (memory.max, cpuset.cpus) when the daemon may
+  write /sys/fs/cgroup;
+* volumes as directories "bound" into the rootfs by symlink.
+
+This is what runs on dockerd-less MI355X boxes (gpurun, bench.py): the
+control-plane latency being measured — schedule → materialize → spawn →
+running — is real work, not a stub.
+
+Processes are stopped by exact PID/process-group only (never by pattern).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import os
+import shutil
+import signal
+import subprocess
+import time
+import uuid as uuidlib
+from typing import Dict, List, Optional
+
+from ..models.etcd import ContainerSpec
+from ..xerrors import ContainerExisted, ContainerNotExist, VolumeExisted
+from .base import ContainerState, GpuResolver, RuntimeDriver, VolumeState
+from .devices import visible_device_env
+
+CGROUP_ROOT = "/sys/fs/cgroup"
+DEFAULT_CMD = ["sleep", "infinity"]
+
+
+class _Proc:
+    def __init__(self, state: ContainerState, spec: ContainerSpec) -> None:
+        self.state = state
+        self.spec = spec
+        self.popen: Optional[subprocess.Popen] = None
+        self.cgroup: Optional[str] = None
+
+
+class ProcRuntime(RuntimeDriver):
+    def __init__(
+        self,
+        base_dir: str = "",
+        gpu_resolver: Optional[GpuResolver] = None,
+        use_cgroups: bool = True,
+    ) -> None:
+        self.base = base_dir or os.path.join(os.getcwd(), ".state", "procrt")
+        for sub in ("containers", "volumes", "images"):
+            os.makedirs(os.path.join(self.base, sub), exist_ok=True)
+        self.gpu_resolver: GpuResolver = gpu_resolver or (lambda _u: None)
+        self.use_cgroups = use_cgroups
+        self._procs: Dict[str, _Proc] = {}
+        self.volumes: Dict[str, VolumeState] = {}
+        self._load_volumes()
+
+    # ------------------------------------------------------------------ util
+    def _cdir(self, name: str) -> str:
+        return os.path.join(self.base, "containers", name)
+
+    def _image_dir(self, ref: str) -> str:
+        return os.path.join(self.base, "images", ref.replace("/", "_").replace(":", "_"))
+
+    def _load_volumes(self) -> None:
+        vroot = os.path.join(self.base, "volumes")
+        for name in os.listdir(vroot):
+            mp = os.path.join(vroot, name, "_data")
+            optf = os.path.join(vroot, name, "opts.json")
+            opts = {}
+            if os.path.exists(optf):
+                try:
+                    opts = json.load(open(optf))
+                except Exception:
+                    opts = {}
+            if os.path.isdir(mp):
+                self.volumes[name] = VolumeState(name=name, mountpoint=mp, options=opts)
+
+    def _refresh(self, p: _Proc) -> None:
+        if p.popen is not None and p.popen.poll() is not None and p.state.running:
+            p.state.running = False
+            p.state.status = "exited"
+            p.state.pid = 0
+
+    # ------------------------------------------------------------ containers
+    async def create(self, spec: ContainerSpec) -> str:
+        name = spec.container_name
+        if name in self._procs:
+            raise ContainerExisted(name)
+        cid = uuidlib.uuid4().hex[:12]
+        rootfs = os.path.join(self._cdir(name), "rootfs")
+        os.makedirs(rootfs, exist_ok=True)
+        seed = self._image_dir(spec.image) if spec.image else ""
+        if seed and os.path.isdir(seed):
+            shutil.copytree(seed, rootfs, dirs_exist_ok=True)
+        # bind volumes/host dirs into the rootfs by symlink
+        binds = list(spec.host_config.get("Binds") or [])
+        for b in binds:
+            src, _, dest = b.partition(":")
+            dest = dest.split(":")[0]
+            if not src or not dest:
+                continue
+            vol = self.volumes.get(src)
+            src_path = vol.mountpoint if vol else src
+            link = os.path.join(rootfs, dest.lstrip("/"))
+            os.makedirs(os.path.dirname(link), exist_ok=True)
+            if not os.path.lexists(link):
+                os.symlink(src_path, link)
+        st = ContainerState(
+            id=cid,
+            name=name,
+            image=spec.image,
+            status="created",
+            env=list(spec.env),
+            gpu_uuids=list(spec.gpu_uuids),
+            cpuset_cpus=spec.cpuset_cpus,
+            memory=spec.memory_bytes,
+            port_bindings=dict(spec.host_config.get("PortBindings") or {}),
+            upper_dir=rootfs,
+            binds=binds,
+        )
+        p = _Proc(st, spec)
+        self._procs[name] = p
+        with open(os.path.join(self._cdir(name), "spec.json"), "w") as f:
+            f.write(spec.serialize())
+        return cid
+
+    def _get(self, name: str) -> _Proc:
+        if name not in self._procs:
+            raise ContainerNotExist(name)
+        return self._procs[name]
+
+    def _env_for(self, p: _Proc) -> Dict[str, str]:
+        env = dict(os.environ)
+        env.update(visible_device_env(p.state.gpu_uuids, self.gpu_resolver))
+        for e in p.state.env:
+            k, _, v = e.partition("=")
+            env[k] = v
+        return env
+
+    def _setup_cgroup(self, name: str, p: _Proc, pid: int) -> None:
+        if not self.use_cgroups:
+            return
+        cg = os.path.join(CGROUP_ROOT, "gda", name)
+        try:
+            os.makedirs(cg, exist_ok=True)
+            if p.state.memory > 0:
+                with open(os.path.join(cg, "memory.max"), "w") as f:
+                    f.write(str(p.state.memory))
+            if p.state.cpuset_cpus:
+                with open(os.path.join(cg, "cpuset.cpus"), "w") as f:
+                    f.write(p.state.cpuset_cpus)
+            with open(os.path.join(cg, "cgroup.procs"), "w") as f:
+                f.write(str(pid))
+            p.cgroup = cg
+        except OSError:
+            p.cgroup = None  # best-effort: not permitted in this environment
+
+    async def start(self, name: str) -> None:
+        p = self._get(name)
+        self._refresh(p)
+        if p.state.running:
+            return
+        cmd = list(p.spec.config.get("Cmd") or []) or DEFAULT_CMD
+        logf = open(os.path.join(self._cdir(name), "console.log"), "ab")
+        try:
+            p.popen = subprocess.Popen(
+                cmd,
+                cwd=p.state.upper_dir,
+                env=self._env_for(p),
+                stdout=logf,
+                stderr=subprocess.STDOUT,
+                start_new_session=True,  # own pgid: exact-target signalling
+            )
+        finally:
+            logf.close()
+        p.state.pid = p.popen.pid
+        p.state.running, p.state.paused, p.state.status = True, False, "running"
+        self._setup_cgroup(name, p, p.popen.pid)
+
+    def _signal_group(self, p: _Proc, sig: int) -> None:
+        if p.popen is None or p.popen.poll() is not None:
+            return
+        try:
+            os.killpg(os.getpgid(p.popen.pid), sig)
+        except ProcessLookupError:
+            pass
+
+    async def stop(self, name: str, timeout: int = 10) -> None:
+        p = self._get(name)
+        self._refresh(p)
+        if p.popen is not None and p.popen.poll() is None:
+            self._signal_group(p, signal.SIGTERM)
+            deadline = time.monotonic() + timeout
+            while time.monotonic() < deadline and p.popen.poll() is None:
+                await asyncio.sleep(0.01)
+            if p.popen.poll() is None:
+                self._signal_group(p, signal.SIGKILL)
+                p.popen.wait(timeout=5)
+        p.state.running, p.state.paused, p.state.status = False, False, "exited"
+        p.state.pid = 0
+
+    async def pause(self, name: str) -> None:
+        p = self._get(name)
+        if p.cgroup:
+            try:
+                with open(os.path.join(p.cgroup, "cgroup.freeze"), "w") as f:
+                    f.write("1")
+                p.state.paused, p.state.status = True, "paused"
+                return
+            except OSError:
+                pass
+        self._signal_group(p, signal.SIGSTOP)
+        p.state.paused, p.state.status = True, "paused"
+
+    async def unpause(self, name: str) -> None:
+        p = self._get(name)
+        if p.cgroup:
+            try:
+                with open(os.path.join(p.cgroup, "cgroup.freeze"), "w") as f:
+                    f.write("0")
+                p.state.paused, p.state.status = False, "running"
+                return
+            except OSError:
+                pass
+        self._signal_group(p, signal.SIGCONT)
+        p.state.paused, p.state.status = False, "running"
+
+    async def restart(self, name: str, timeout: int = 10) -> None:
+        await self.stop(name, timeout)
+        await self.start(name)
+
+    async def remove(self, name: str, force: bool = True) -> None:
+        p = self._get(name)
+        self._refresh(p)
+        if p.state.running:
+            if not force:
+                raise RuntimeError(f"{name} is running")
+            await self.stop(name, timeout=2)
+        if p.cgroup:
+            try:
+                os.rmdir(p.cgroup)
+            except OSError:
+                pass
+        self._procs.pop(name, None)
+        shutil.rmtree(self._cdir(name), ignore_errors=True)
+
+    async def inspect(self, name: str) -> Optional[ContainerState]:
+        p = self._procs.get(name)
+        if p is None:
+            return None
+        self._refresh(p)
+        return p.state
+
+    async def list(self, all: bool = True) -> List[ContainerState]:
+        for p in self._procs.values():
+            self._refresh(p)
+        return [p.state for p in self._procs.values() if all or p.state.running]
+
+    async def execute(self, name: str, cmd: List[str], workdir: str = "") -> str:
+        p = self._get(name)
+        self._refresh(p)
+        if not p.state.running:
+            raise RuntimeError(f"{name} is not running")
+        cwd = (
+            os.path.join(p.state.upper_dir, workdir.lstrip("/"))
+            if workdir
+            else p.state.upper_dir
+        )
+        os.makedirs(cwd, exist_ok=True)
+        proc = await asyncio.create_subprocess_exec(
+            *cmd,
+            cwd=cwd,
+            env=self._env_for(p),
+            stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT,
+        )
+        out, _ = await proc.communicate()
+        return out.decode(errors="replace")
+
+    async def commit(self, name: str, image: str, tag: str = "") -> str:
+        p = self._get(name)
+        ref = f"{image}:{tag}" if tag else image
+        dest = self._image_dir(ref)
+        shutil.rmtree(dest, ignore_errors=True)
+        shutil.copytree(p.state.upper_dir, dest, symlinks=True)
+        return ref
+
+    # --------------------------------------------------------------- volumes
+    async def volume_create(
+        self, name: str, driver_opts: Optional[Dict[str, str]] = None
+    ) -> VolumeState:
+        if name in self.volumes:
+            raise VolumeExisted(name)
+        vdir = os.path.join(self.base, "volumes", name)
+        mp = os.path.join(vdir, "_data")
+        os.makedirs(mp, exist_ok=True)
+        opts = dict(driver_opts or {})
+        with open(os.path.join(vdir, "opts.json"), "w") as f:
+            json.dump(opts, f)
+        vs = VolumeState(name=name, mountpoint=mp, options=opts)
+        self.volumes[name] = vs
+        return vs
+
+    async def volume_remove(self, name: str, force: bool = True) -> None:
+        self.volumes.pop(name, None)
+        shutil.rmtree(os.path.join(self.base, "volumes", name), ignore_errors=True)
+
+    async def volume_inspect(self, name: str) -> Optional[VolumeState]:
+        return self.volumes.get(name)
+
+    async def close(self) -> None:
+        for name in list(self._procs):
+            p = self._procs[name]
+            if p.popen is not None and p.popen.poll() is None:
+                await self.stop(name, timeout=2)
