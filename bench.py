@@ -1,0 +1,227 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: the BASELINE.json headline metric.
+
+Measures p50 replicaSet create->running + patch(GPU re-scale) turnaround
+through the full control plane (HTTP daemon + schedulers + state store +
+proc runtime with real processes and real GPU visibility injection).
+
+Topology: ONE daemon owns the node's GPUs (rank 0 serves it on 127.0.0.1);
+all N ranks are concurrent tenants, each driving its own replicaSet
+lifecycle cycle (create 1-GPU -> running, patch gpuCount 1->0 = rolling
+replace with GPU re-schedule, delete). Weak scaling: per-rank work is fixed
+as N grows; N concurrent 1-GPU replicaSets saturate N GPUs (BASELINE
+configs #2 and #4 combined).
+
+Usage (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  # N>1 is launched via torch.distributed.run; RANK/WORLD_SIZE read from env
+
+Rank 0 prints ONE JSON line with the aggregate metric.
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import statistics
+import threading
+import time
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=32)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--port", type=int, default=0, help="daemon port (default MASTER_PORT+1711 or 18731)")
+    return p.parse_args()
+
+
+def dist_env():
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    return world, rank, local_rank
+
+
+def daemon_port(args) -> int:
+    if args.port:
+        return args.port
+    base = int(os.environ.get("MASTER_PORT", "17020"))
+    return base + 1711
+
+
+class DaemonThread:
+    """Runs the control-plane daemon + uvicorn in a background thread."""
+
+    def __init__(self, port: int, n_gpus: int, use_gpu: bool, data_dir: str):
+        from gpu_docker_api_amd.config import Config
+
+        self.cfg = Config(
+            addr=f"127.0.0.1:{port}",
+            state="memory",
+            data_dir=data_dir,
+            runtime="proc",
+            inventory="auto" if use_gpu else "mock",
+            mock_gpus=max(8, n_gpus),
+            copy_engine="auto",
+            run_xgmi_probe=use_gpu,  # native HIP probe, outside the timed region
+            port_range="41000-42000",
+        )
+        self.port = port
+        self._thread: threading.Thread | None = None
+        self._server = None
+
+    def start(self):
+        import uvicorn
+
+        from gpu_docker_api_amd.routers.app import build_app
+
+        app = build_app(self.cfg)
+        config = uvicorn.Config(app, host="127.0.0.1", port=self.port, log_level="warning")
+        self._server = uvicorn.Server(config)
+        self._thread = threading.Thread(target=self._server.run, daemon=True)
+        self._thread.start()
+        # wait for readiness
+        import httpx
+
+        deadline = time.time() + 180
+        while time.time() < deadline:
+            try:
+                if httpx.get(f"http://127.0.0.1:{self.port}/ping", timeout=2).status_code == 200:
+                    return
+            except Exception:
+                time.sleep(0.2)
+        raise RuntimeError("daemon did not become ready")
+
+    def stop(self):
+        if self._server is not None:
+            self._server.should_exit = True
+        if self._thread is not None:
+            self._thread.join(timeout=30)
+
+
+def one_cycle(client, name: str) -> float:
+    """One timed cycle: create(1 GPU)->running, patch gpu 1->0 (rolling
+    replace), delete. Returns latency in ms. Raises on any non-200 code."""
+    t0 = time.perf_counter()
+    r = client.post(
+        "/api/v1/replicaSet",
+        json={
+            "imageName": "synthetic:bench",
+            "replicaSetName": name,
+            "gpuCount": 1,
+            "cpuCount": 1,
+            "memory": "1GB",
+        },
+    ).json()
+    assert r["code"] == 200, f"run failed: {r}"
+    r = client.patch(f"/api/v1/replicaSet/{name}", json={"gpuPatch": {"gpuCount": 0}}).json()
+    assert r["code"] == 200, f"patch failed: {r}"
+    r = client.delete(f"/api/v1/replicaSet/{name}").json()
+    assert r["code"] == 200, f"delete failed: {r}"
+    return (time.perf_counter() - t0) * 1000.0
+
+
+def main():
+    args = parse_args()
+    world, rank, local_rank = dist_env()
+    n_gpus = max(args.gpus, world)
+
+    import torch
+
+    use_gpu = torch.cuda.is_available()
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if use_gpu else "gloo"
+        dist.init_process_group(backend=backend)
+        if use_gpu:
+            torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    port = daemon_port(args)
+    server = None
+    if rank == 0:
+        server = DaemonThread(
+            port, n_gpus, use_gpu, data_dir=os.path.join("/tmp", f"gda-bench-{port}")
+        )
+        import shutil
+
+        shutil.rmtree(server.cfg.data_dir, ignore_errors=True)
+        server.start()
+    barrier_sync()
+
+    import httpx
+
+    client = httpx.Client(base_url=f"http://127.0.0.1:{port}", timeout=120)
+    name = f"bench{rank}"
+    # make sure a previous run's leftovers are gone
+    client.delete(f"/api/v1/replicaSet/{name}")
+
+    for _ in range(args.warmup):
+        one_cycle(client, name)
+
+    barrier_sync()
+    t_start = time.perf_counter()
+    latencies = [one_cycle(client, name) for _ in range(args.steps)]
+    barrier_sync()
+    elapsed_s = time.perf_counter() - t_start
+
+    # aggregate across ranks
+    if dist is not None:
+        all_lat: list = [None] * world
+        all_elapsed: list = [None] * world
+        dist.all_gather_object(all_lat, latencies)
+        dist.all_gather_object(all_elapsed, elapsed_s)
+    else:
+        all_lat = [latencies]
+        all_elapsed = [elapsed_s]
+
+    if rank == 0:
+        merged = sorted(x for l in all_lat for x in l)
+        p50 = statistics.median(merged)
+        max_elapsed = max(all_elapsed)
+        result = {
+            "metric": "p50 replicaSet create->running + patch(GPU rescale) latency",
+            "value": round(p50, 3),
+            "unit": "ms",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(max_elapsed * 1000.0 / args.steps, 3),
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,  # the reference publishes no numbers
+            "dtype": "n/a",
+            "data": "synthetic containers (proc runtime, per-rank 1-GPU lifecycle)",
+            "config": {
+                "model": "replicaSet lifecycle: create(1 GPU)->running, patch gpu 1->0 rolling replace, delete",
+                "global_batch": world,
+                "seq_len": 0,
+                "parallelism": f"{world} concurrent tenants, 1 daemon, {n_gpus} GPUs",
+                "runtime": "proc",
+                "throughput_cycles_per_s": round(world * args.steps / max_elapsed, 3),
+                "p95_ms": round(merged[min(int(0.95 * len(merged)), len(merged) - 1)], 3),
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    client.close()
+    if dist is not None:
+        dist.barrier()
+        dist.destroy_process_group()
+    if server is not None:
+        server.stop()
+
+
+if __name__ == "__main__":
+    main()

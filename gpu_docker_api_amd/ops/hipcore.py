@@ -1,0 +1,117 @@
+"""Python facade over the _hipcore extension + the rccl_smoke binary.
+
+Fails loudly when a GPU is present but the native extension is missing
+(NativeOpUnavailable) — GPU paths must never silently fall back.
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import os
+import subprocess
+from typing import Optional
+
+from ..xerrors import NativeOpUnavailable
+
+_ext = None
+_ext_err: Optional[str] = None
+
+
+def load_ext():
+    """Load the _hipcore torch extension (torch must be imported first so
+    libtorch*.so are resolvable)."""
+    global _ext, _ext_err
+    if _ext is not None:
+        return _ext
+    if _ext_err is not None:
+        raise NativeOpUnavailable(_ext_err)
+    try:
+        import importlib.util
+        import torch  # noqa: F401  (loads libtorch into the process)
+
+        so = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_hipcore.so")
+        if not os.path.exists(so):
+            raise FileNotFoundError(f"{so} not built (run ops.build)")
+        spec = importlib.util.spec_from_file_location("_hipcore", so)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        _ext = mod
+        return _ext
+    except Exception as exc:
+        _ext_err = f"native _hipcore unavailable: {exc}"
+        raise NativeOpUnavailable(_ext_err) from exc
+
+
+def gpu_available() -> bool:
+    try:
+        import torch
+
+        return torch.cuda.is_available()
+    except Exception:
+        return False
+
+
+def run_probe(mib: int = 512, iters: int = 5) -> dict:
+    """Measure HBM stream + pairwise p2p bandwidth across visible GPUs.
+
+    Returns {"gpus": [uuid...], "hbm_gbps": [...], "p2p_gbps": [[...]]} —
+    the shape parallel.topology.Topology.overlay_measured consumes. GPU
+    identity comes from torch (HIP enumeration order); UUIDs are taken from
+    the amdsmi inventory at matching indices when available.
+    """
+    ext = load_ext()
+    n = ext.device_count()
+    if n == 0:
+        raise NativeOpUnavailable("no HIP devices visible")
+    uuids = [f"GPU-{i}" for i in range(n)]
+    try:
+        from ..parallel.inventory import AmdSmiInventory
+
+        infos = AmdSmiInventory().enumerate()
+        if len(infos) == n:
+            uuids = [g.uuid for g in infos]
+    except Exception:
+        pass
+    hbm = [round(ext.stream_bandwidth_gbps(i, mib, iters), 1) for i in range(n)]
+    p2p = [[0.0] * n for _ in range(n)]
+    for i in range(n):
+        for j in range(n):
+            if i != j:
+                p2p[i][j] = round(ext.p2p_bandwidth_gbps(i, j, mib, iters), 1)
+    return {"gpus": uuids, "hbm_gbps": hbm, "p2p_gbps": p2p}
+
+
+async def run_probe_async(mib: int = 512, iters: int = 5) -> dict:
+    return await asyncio.get_running_loop().run_in_executor(
+        None, lambda: run_probe(mib, iters)
+    )
+
+
+def rccl_smoke_path() -> str:
+    root = os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    return os.path.join(root, "csrc", "bin", "rccl_smoke")
+
+
+def run_rccl_smoke(ndev: int = 0, mib: int = 64, timeout: float = 120.0) -> dict:
+    """Run the native RCCL all-reduce smoke binary; returns its JSON verdict."""
+    binary = rccl_smoke_path()
+    if not os.path.exists(binary):
+        raise NativeOpUnavailable(f"{binary} not built (run ops.build)")
+    cmd = [binary]
+    if ndev > 0:
+        cmd.append(str(ndev))
+        cmd.append(str(mib))
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=timeout)
+    line = (out.stdout.strip().splitlines() or ["{}"])[-1]
+    try:
+        result = json.loads(line)
+    except json.JSONDecodeError:
+        result = {"ok": False, "error": f"unparseable output: {line!r}", "rc": out.returncode}
+    result.setdefault("ok", False)
+    return result
+
+
+async def run_rccl_smoke_async(ndev: int = 0, mib: int = 64) -> dict:
+    return await asyncio.get_running_loop().run_in_executor(
+        None, lambda: run_rccl_smoke(ndev, mib)
+    )
