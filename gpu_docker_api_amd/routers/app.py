@@ -43,9 +43,9 @@ log = logging.getLogger(__name__)
 class Daemon:
     """Holds every initialized subsystem; built once per process."""
 
-    def __init__(self, cfg: Config) -> None:
+    def __init__(self, cfg: Config, store: Optional[StateStore] = None) -> None:
         self.cfg = cfg
-        self.store: Optional[StateStore] = None
+        self.store: Optional[StateStore] = store
         self.queue: Optional[WorkQueue] = None
         self.gpu: Optional[GpuScheduler] = None
         self.cpu: Optional[CpuScheduler] = None
@@ -62,12 +62,15 @@ class Daemon:
         os.makedirs(cfg.data_dir, exist_ok=True)
         os.makedirs(cfg.merges_dir, exist_ok=True)
 
-        if cfg.state.startswith("etcd:"):
-            self.store = EtcdGatewayStore(cfg.state.split(":", 1)[1] or "http://127.0.0.1:2379")
-        elif cfg.state == "memory":
-            self.store = MemoryStore(wal_path=cfg.wal_path)
-        else:
-            self.store = MemoryStore()
+        if self.store is None:
+            if cfg.state.startswith("etcd:"):
+                self.store = EtcdGatewayStore(
+                    cfg.state.split(":", 1)[1] or "http://127.0.0.1:2379"
+                )
+            elif cfg.state == "memory":
+                self.store = MemoryStore(wal_path=cfg.wal_path)
+            else:
+                self.store = MemoryStore()
         self.queue = WorkQueue(self.store)
         self.queue.start()
 

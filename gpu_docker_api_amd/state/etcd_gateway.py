@@ -54,10 +54,17 @@ def _kv_from_json(kv: Dict[str, Any]) -> KeyValue:
 
 
 class EtcdGatewayStore(StateStore):
-    def __init__(self, endpoint: str = "http://127.0.0.1:2379", timeout: float = 2.0) -> None:
+    def __init__(
+        self,
+        endpoint: str = "http://127.0.0.1:2379",
+        timeout: float = 2.0,
+        transport: Optional[httpx.AsyncBaseTransport] = None,
+    ) -> None:
         # reference: dial timeout 2s (etcd/client.go:17), op timeout 1s (common.go:31)
         self.endpoint = endpoint.rstrip("/")
-        self._client = httpx.AsyncClient(base_url=self.endpoint, timeout=timeout)
+        self._client = httpx.AsyncClient(
+            base_url=self.endpoint, timeout=timeout, transport=transport
+        )
 
     async def _call(self, path: str, body: Dict[str, Any]) -> Dict[str, Any]:
         resp = await self._client.post(path, json=body)

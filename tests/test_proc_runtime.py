@@ -1,0 +1,153 @@
+"""ProcRuntime: real supervised processes, no GPU required (GPU visibility
+injection is covered by test_gpu.py on the MI355X box)."""
+import os
+import signal
+import time
+
+from gpu_docker_api_amd.models.etcd import ContainerSpec
+from gpu_docker_api_amd.runtime.proc import ProcRuntime
+
+
+def _spec(name, cmd=None, env=None, image=""):
+    s = ContainerSpec()
+    s.config = {"Image": image, "Env": list(env or []), "Cmd": list(cmd or [])}
+    s.host_config = {}
+    s.container_name = name
+    return s
+
+
+def test_lifecycle_real_process(tmp_path, run):
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        await rt.create(_spec("c-1", cmd=["sleep", "30"]))
+        st = await rt.inspect("c-1")
+        assert st is not None and not st.running
+        await rt.start("c-1")
+        st = await rt.inspect("c-1")
+        assert st.running and st.pid > 0
+        os.kill(st.pid, 0)  # process really exists
+        await rt.stop("c-1")
+        st = await rt.inspect("c-1")
+        assert not st.running and st.pid == 0
+        await rt.start("c-1")
+        assert (await rt.inspect("c-1")).running
+        await rt.remove("c-1", force=True)
+        assert await rt.inspect("c-1") is None
+        await rt.close()
+
+    run(main())
+
+
+def test_exited_process_detected(tmp_path, run):
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        await rt.create(_spec("c-1", cmd=["sh", "-c", "exit 0"]))
+        await rt.start("c-1")
+        for _ in range(100):
+            st = await rt.inspect("c-1")
+            if not st.running:
+                break
+            time.sleep(0.05)
+        assert not st.running and st.status == "exited"
+        await rt.close()
+
+    run(main())
+
+
+def test_exec_runs_in_rootfs_with_env(tmp_path, run):
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        await rt.create(_spec("c-1", cmd=["sleep", "30"], env=["FOO=bar42"]))
+        await rt.start("c-1")
+        out = await rt.execute("c-1", ["sh", "-c", "echo $FOO; pwd"])
+        assert "bar42" in out
+        st = await rt.inspect("c-1")
+        assert st.upper_dir in out
+        # workdir
+        out = await rt.execute("c-1", ["pwd"], workdir="/sub/dir")
+        assert out.strip().endswith("sub/dir")
+        await rt.close()
+
+    run(main())
+
+
+def test_pause_unpause_signals(tmp_path, run):
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        await rt.create(_spec("c-1", cmd=["sleep", "30"]))
+        await rt.start("c-1")
+        st = await rt.inspect("c-1")
+        await rt.pause("c-1")
+        # process group should be SIGSTOPped
+        with open(f"/proc/{st.pid}/stat") as f:
+            state = f.read().split(") ")[1].split()[0]
+        assert state in ("T", "t"), f"expected stopped, got {state}"
+        assert (await rt.inspect("c-1")).paused
+        await rt.unpause("c-1")
+        with open(f"/proc/{st.pid}/stat") as f:
+            state = f.read().split(") ")[1].split()[0]
+        assert state in ("S", "R"), f"expected running, got {state}"
+        await rt.close()
+
+    run(main())
+
+
+def test_commit_and_image_seed(tmp_path, run):
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        await rt.create(_spec("c-1", cmd=["sleep", "30"]))
+        await rt.start("c-1")
+        st = await rt.inspect("c-1")
+        with open(os.path.join(st.upper_dir, "state.txt"), "w") as f:
+            f.write("snapshot-me")
+        ref = await rt.commit("c-1", "myimg", "v1")
+        assert ref == "myimg:v1"
+        # new container from the committed image inherits the rootfs
+        await rt.create(_spec("c2-1", cmd=["sleep", "30"], image="myimg:v1"))
+        st2 = await rt.inspect("c2-1")
+        assert open(os.path.join(st2.upper_dir, "state.txt")).read() == "snapshot-me"
+        await rt.close()
+
+    run(main())
+
+
+def test_volume_bind_symlink(tmp_path, run):
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        vol = await rt.volume_create("v-1", {"size": "1GB"})
+        with open(os.path.join(vol.mountpoint, "data.txt"), "w") as f:
+            f.write("vol-data")
+        spec = _spec("c-1", cmd=["sleep", "30"])
+        spec.host_config["Binds"] = ["v-1:/data"]
+        await rt.create(spec)
+        await rt.start("c-1")
+        out = await rt.execute("c-1", ["cat", "data.txt"], workdir="/data")
+        assert "vol-data" in out
+        # volumes survive runtime restart (opts.json reload)
+        rt2 = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        v = await rt2.volume_inspect("v-1")
+        assert v is not None and v.options.get("size") == "1GB"
+        await rt.close()
+
+    run(main())
+
+
+def test_stop_kills_whole_process_group(tmp_path, run):
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        # parent spawns a child; stop must kill both (exact pgid, no patterns)
+        await rt.create(_spec("c-1", cmd=["sh", "-c", "sleep 60 & wait"]))
+        await rt.start("c-1")
+        st = await rt.inspect("c-1")
+        pgid = os.getpgid(st.pid)
+        await rt.stop("c-1", timeout=3)
+        time.sleep(0.1)
+        try:
+            os.killpg(pgid, 0)
+            alive = True
+        except ProcessLookupError:
+            alive = False
+        assert not alive, "process group survived stop"
+        await rt.close()
+
+    run(main())
