@@ -1,44 +1,81 @@
 """CORS + static-bearer auth middleware (reference: routers/cors.go:9-32,
-routers/auth.go:9-26 — auth disabled when APIKEY env is unset)."""
+routers/auth.go:9-26 — auth disabled when APIKEY env is unset).
+
+Implemented as raw ASGI wrappers, not starlette BaseHTTPMiddleware: the
+latter re-buffers every request/response through an ASGI sub-app and costs
+~1-2 ms per request — measurable against this daemon's ~1-4 ms operations.
+"""
 from __future__ import annotations
 
-from starlette.middleware.base import BaseHTTPMiddleware
-from starlette.requests import Request
+import json
 
-from .codes import Code
-from .response import error
-
-
-class CorsMiddleware(BaseHTTPMiddleware):
-    async def dispatch(self, request: Request, call_next):
-        origin = request.headers.get("origin", "*")
-        if request.method == "OPTIONS":
-            from starlette.responses import Response
-
-            resp = Response(status_code=204)
-        else:
-            resp = await call_next(request)
-        resp.headers["Access-Control-Allow-Origin"] = origin
-        resp.headers["Access-Control-Allow-Credentials"] = "true"
-        resp.headers["Access-Control-Allow-Headers"] = (
-            "Content-Type, Content-Length, Authorization, Origin, X-Requested-With"
-        )
-        resp.headers["Access-Control-Allow-Methods"] = (
-            "GET, POST, PUT, PATCH, DELETE, OPTIONS"
-        )
-        return resp
+_CORS_HEADERS = [
+    (b"access-control-allow-credentials", b"true"),
+    (
+        b"access-control-allow-headers",
+        b"Content-Type, Content-Length, Authorization, Origin, X-Requested-With",
+    ),
+    (b"access-control-allow-methods", b"GET, POST, PUT, PATCH, DELETE, OPTIONS"),
+]
 
 
-class AuthMiddleware(BaseHTTPMiddleware):
+class CorsMiddleware:
+    def __init__(self, app) -> None:
+        self.app = app
+
+    async def __call__(self, scope, receive, send):
+        if scope["type"] != "http":
+            return await self.app(scope, receive, send)
+        origin = b"*"
+        for k, v in scope.get("headers", []):
+            if k == b"origin":
+                origin = v
+                break
+        extra = [(b"access-control-allow-origin", origin)] + _CORS_HEADERS
+
+        if scope.get("method") == "OPTIONS":
+            await send(
+                {"type": "http.response.start", "status": 204, "headers": extra}
+            )
+            await send({"type": "http.response.body", "body": b""})
+            return
+
+        async def send_with_cors(message):
+            if message["type"] == "http.response.start":
+                message = dict(message)
+                message["headers"] = list(message.get("headers", [])) + extra
+            await send(message)
+
+        await self.app(scope, receive, send_with_cors)
+
+
+class AuthMiddleware:
     def __init__(self, app, apikey: str = "") -> None:
-        super().__init__(app)
-        self.apikey = apikey
+        self.app = app
+        self.apikey = apikey.encode()
 
-    async def dispatch(self, request: Request, call_next):
-        if not self.apikey or request.url.path in ("/ping", "/metrics"):
-            return await call_next(request)
-        auth = request.headers.get("authorization", "")
-        token = auth[7:] if auth.lower().startswith("bearer ") else auth
+    async def __call__(self, scope, receive, send):
+        if scope["type"] != "http" or not self.apikey:
+            return await self.app(scope, receive, send)
+        path = scope.get("path", "")
+        if path in ("/ping", "/metrics"):
+            return await self.app(scope, receive, send)
+        token = b""
+        for k, v in scope.get("headers", []):
+            if k == b"authorization":
+                token = v[7:] if v[:7].lower() == b"bearer " else v
+                break
         if token != self.apikey:
-            return error(Code.FORBIDDEN)
-        return await call_next(request)
+            body = json.dumps(
+                {"code": 403, "msg": "Forbidden", "data": None}
+            ).encode()
+            await send(
+                {
+                    "type": "http.response.start",
+                    "status": 200,  # envelope always HTTP 200 (reference contract)
+                    "headers": [(b"content-type", b"application/json")],
+                }
+            )
+            await send({"type": "http.response.body", "body": body})
+            return
+        await self.app(scope, receive, send)

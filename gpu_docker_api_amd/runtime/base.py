@@ -58,6 +58,11 @@ class VolumeState:
 class RuntimeDriver:
     """All methods raise on hard failure; inspect returns None when absent."""
 
+    # True when the driver owns the container rootfs directory outright
+    # (proc/mock): the control plane may rename it away during replacement.
+    # False for docker (UpperDir belongs to overlayfs; must be copied).
+    owns_rootfs = False
+
     async def create(self, spec: ContainerSpec) -> str:
         """Create (not start) a container named spec.container_name; returns id."""
         raise NotImplementedError

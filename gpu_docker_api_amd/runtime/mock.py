@@ -23,6 +23,8 @@ from .base import ContainerState, RuntimeDriver, VolumeState
 
 
 class MockRuntime(RuntimeDriver):
+    owns_rootfs = True
+
     def __init__(self, base_dir: str = "", fail_on: Optional[set] = None) -> None:
         self.base = base_dir or os.path.join(os.getcwd(), ".state", "mockrt")
         os.makedirs(os.path.join(self.base, "containers"), exist_ok=True)

@@ -50,6 +50,8 @@ class _Proc:
 
 
 class ProcRuntime(RuntimeDriver):
+    owns_rootfs = True
+
     def __init__(
         self,
         base_dir: str = "",
@@ -202,8 +204,10 @@ class ProcRuntime(RuntimeDriver):
         if p.popen is not None and p.popen.poll() is None:
             self._signal_group(p, signal.SIGTERM)
             deadline = time.monotonic() + timeout
+            delay = 0.0002  # most processes die in <1 ms; back off geometrically
             while time.monotonic() < deadline and p.popen.poll() is None:
-                await asyncio.sleep(0.01)
+                await asyncio.sleep(delay)
+                delay = min(delay * 2, 0.02)
             if p.popen.poll() is None:
                 self._signal_group(p, signal.SIGKILL)
                 p.popen.wait(timeout=5)

@@ -304,15 +304,27 @@ class ReplicaSetService:
         timer.mark("copy")
         await self.runtime.start(vname_new)
         timer.mark("start")
-        # preserve the replaced version's layer for history/rollback
+        # preserve the replaced version's layer for history/rollback.
+        # When the runtime owns the rootfs dir (proc/mock) a rename is enough
+        # — the old container is deleted next anyway; docker's UpperDir
+        # belongs to overlayfs, so it is copied.
         if old_state is not None and old_state.upper_dir and os.path.isdir(old_state.upper_dir):
             merge_path = os.path.join(self.cfg.merges_dir, name, vname_old)
-            os.makedirs(merge_path, exist_ok=True)
             try:
-                await self.copy.copy_dir(old_state.upper_dir, merge_path)
+                if getattr(self.runtime, "owns_rootfs", False):
+                    os.makedirs(os.path.dirname(merge_path), exist_ok=True)
+                    shutil.rmtree(merge_path, ignore_errors=True)
+                    try:
+                        os.rename(old_state.upper_dir, merge_path)
+                    except OSError:  # cross-device: fall back to copying
+                        os.makedirs(merge_path, exist_ok=True)
+                        await self.copy.copy_dir(old_state.upper_dir, merge_path)
+                else:
+                    os.makedirs(merge_path, exist_ok=True)
+                    await self.copy.copy_dir(old_state.upper_dir, merge_path)
+                self.merges.set(vname_old, merge_path)
             except Exception:
                 log.exception("preserving %s layer failed", vname_old)
-            self.merges.set(vname_old, merge_path)
         timer.mark("preserve")
         # delete the old container, releasing its host ports
         if old_state is not None:
