@@ -49,6 +49,25 @@ __global__ void copy_f32x4_kernel(const float4* __restrict__ src,
   for (; i < n4; i += stride) dst[i] = src[i];
 }
 
+// 4x-unrolled variant: four independent dwordx4 loads in flight per thread
+// before any store — deeper MLP to hide HBM latency on large streams.
+__global__ void copy_f32x4_x4_kernel(const float4* __restrict__ src,
+                                     float4* __restrict__ dst, size_t n4) {
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  size_t i = blockIdx.x * (size_t)blockDim.x + threadIdx.x;
+  for (; i + 3 * stride < n4; i += 4 * stride) {
+    float4 a = src[i];
+    float4 b = src[i + stride];
+    float4 c = src[i + 2 * stride];
+    float4 d = src[i + 3 * stride];
+    dst[i] = a;
+    dst[i + stride] = b;
+    dst[i + 2 * stride] = c;
+    dst[i + 3 * stride] = d;
+  }
+  for (; i < n4; i += stride) dst[i] = src[i];
+}
+
 // Tail-safe scalar copy for non-multiple-of-4 sizes.
 __global__ void copy_f32_tail_kernel(const float* __restrict__ src,
                                      float* __restrict__ dst, size_t n,
@@ -63,9 +82,15 @@ void launch_copy_f32(const float* src, float* dst, size_t n, hipStream_t stream)
   if (n4 > 0) {
     // ≫ 256 workgroups to fill 256 CUs across 8 XCDs; cap to keep launch sane
     int grid = (int)std::min<size_t>((n4 + kBlock - 1) / kBlock, 32768);
-    hipLaunchKernelGGL(copy_f32x4_kernel, dim3(grid), dim3(kBlock), 0, stream,
-                       reinterpret_cast<const float4*>(src),
-                       reinterpret_cast<float4*>(dst), n4);
+    if (n4 >= (size_t)4 * grid * kBlock) {
+      hipLaunchKernelGGL(copy_f32x4_x4_kernel, dim3(grid), dim3(kBlock), 0,
+                         stream, reinterpret_cast<const float4*>(src),
+                         reinterpret_cast<float4*>(dst), n4);
+    } else {
+      hipLaunchKernelGGL(copy_f32x4_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                         reinterpret_cast<const float4*>(src),
+                         reinterpret_cast<float4*>(dst), n4);
+    }
   }
   size_t tail = n - n4 * 4;
   if (tail > 0) {
@@ -219,6 +244,25 @@ double stream_bandwidth_gbps(int device, int mib, int iters) {
   return (2.0 * n * sizeof(float)) / (ms * 1e6);
 }
 
+// Reference point: SDMA/blit path via hipMemcpyAsync DtoD on one device.
+double memcpy_bandwidth_gbps(int device, int mib, int iters) {
+  size_t n = (size_t)mib * 1024 * 1024 / sizeof(float);
+  HIP_CHECK(hipSetDevice(device));
+  DeviceBuf src, dst;
+  src.device = dst.device = device;
+  HIP_CHECK(hipMalloc(&src.ptr, n * sizeof(float)));
+  HIP_CHECK(hipMalloc(&dst.ptr, n * sizeof(float)));
+  HIP_CHECK(hipMemset(src.ptr, 1, n * sizeof(float)));
+  double ms = time_kernel_ms(
+      device,
+      [&](hipStream_t s) {
+        HIP_CHECK(hipMemcpyAsync(dst.ptr, src.ptr, n * sizeof(float),
+                                 hipMemcpyDeviceToDevice, s));
+      },
+      iters);
+  return (2.0 * n * sizeof(float)) / (ms * 1e6);
+}
+
 // Peer bandwidth: dst-device kernel pulls from src-device memory (remote
 // reads ride xGMI; per-link peak ≈153 GB/s, 7 links/GPU on an 8-GPU node).
 // Falls back to hipMemcpyPeerAsync (SDMA path) when kernel p2p access is
@@ -364,6 +408,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_warmup", &mfma_warmup, py::arg("device") = 0,
         py::arg("spins") = 20000);
   m.def("stream_bandwidth_gbps", &stream_bandwidth_gbps, py::arg("device") = 0,
+        py::arg("mib") = 1024, py::arg("iters") = 10,
+        py::call_guard<py::gil_scoped_release>());
+  m.def("memcpy_bandwidth_gbps", &memcpy_bandwidth_gbps, py::arg("device") = 0,
         py::arg("mib") = 1024, py::arg("iters") = 10,
         py::call_guard<py::gil_scoped_release>());
   m.def("p2p_bandwidth_gbps", &p2p_bandwidth_gbps, py::arg("src_dev"),
