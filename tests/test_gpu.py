@@ -105,6 +105,21 @@ def test_rccl_smoke_binary():
     assert res["world"] >= 1
 
 
+def test_native_iocopy_loaded_on_gpu_box(tmp_path):
+    """The io_uring engine must be present and active on GPU boxes (no
+    silent tar fallback in production paths)."""
+    require_gpu()
+    from gpu_docker_api_amd.ops import iocopy
+
+    assert iocopy.uring_available(), "io_uring unavailable on the GPU box"
+    src = tmp_path / "src"
+    src.mkdir()
+    (src / "f.bin").write_bytes(b"x" * 123456)
+    stats = iocopy.copy_tree(str(src), str(tmp_path / "dst"))
+    assert stats["io_uring"] is True
+    assert (tmp_path / "dst" / "f.bin").read_bytes() == b"x" * 123456
+
+
 def test_amdsmi_inventory_real():
     require_gpu()
     from gpu_docker_api_amd.parallel.inventory import AmdSmiInventory
