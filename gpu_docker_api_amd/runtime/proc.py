@@ -47,6 +47,14 @@ class _Proc:
         self.spec = spec
         self.popen: Optional[subprocess.Popen] = None
         self.cgroup: Optional[str] = None
+        # restart policy (reference containers run unless-stopped,
+        # services/replicaset.go:73-75; dockerd enforces it there — here the
+        # supervisor task does)
+        self.restart_policy: str = (
+            (spec.host_config.get("RestartPolicy") or {}).get("Name", "") or "no"
+        )
+        self.manually_stopped = False
+        self.restarts = 0
 
 
 class ProcRuntime(RuntimeDriver):
@@ -65,7 +73,36 @@ class ProcRuntime(RuntimeDriver):
         self.use_cgroups = use_cgroups
         self._procs: Dict[str, _Proc] = {}
         self.volumes: Dict[str, VolumeState] = {}
+        self._supervisor: Optional[asyncio.Task] = None
         self._load_volumes()
+
+    # -------------------------------------------------------------- supervisor
+    def _ensure_supervisor(self) -> None:
+        if self._supervisor is None or self._supervisor.done():
+            self._supervisor = asyncio.get_running_loop().create_task(
+                self._supervise_loop()
+            )
+
+    async def _supervise_loop(self) -> None:
+        """Enforce restart policies: a container whose process died (and was
+        not stopped through the API) is restarted when its policy is
+        'always' or 'unless-stopped'."""
+        while True:
+            await asyncio.sleep(0.2)
+            for name, p in list(self._procs.items()):
+                self._refresh(p)
+                if (
+                    not p.state.running
+                    and p.state.status == "exited"
+                    and not p.manually_stopped
+                    and p.restart_policy in ("always", "unless-stopped")
+                    and p.popen is not None  # was started at least once
+                ):
+                    try:
+                        p.restarts += 1
+                        await self.start(name)
+                    except Exception:  # noqa: BLE001 — keep supervising
+                        pass
 
     # ------------------------------------------------------------------ util
     def _cdir(self, name: str) -> str:
@@ -104,7 +141,10 @@ class ProcRuntime(RuntimeDriver):
         os.makedirs(rootfs, exist_ok=True)
         seed = self._image_dir(spec.image) if spec.image else ""
         if seed and os.path.isdir(seed):
-            shutil.copytree(seed, rootfs, dirs_exist_ok=True)
+            # image materialization can be large: off the event loop
+            await asyncio.get_running_loop().run_in_executor(
+                None, lambda: shutil.copytree(seed, rootfs, dirs_exist_ok=True)
+            )
         # bind volumes/host dirs into the rootfs by symlink
         binds = list(spec.host_config.get("Binds") or [])
         for b in binds:
@@ -171,9 +211,14 @@ class ProcRuntime(RuntimeDriver):
     async def start(self, name: str) -> None:
         p = self._get(name)
         self._refresh(p)
+        p.manually_stopped = False
+        self._ensure_supervisor()
         if p.state.running:
             return
         cmd = list(p.spec.config.get("Cmd") or []) or DEFAULT_CMD
+        # spawn inline: fork+exec is ~1 ms, and forking from an executor
+        # thread measurably REGRESSES multi-tenant throughput (fork while
+        # pool threads hold allocator locks) — measured 109 -> 71 cycles/s
         logf = open(os.path.join(self._cdir(name), "console.log"), "ab")
         try:
             p.popen = subprocess.Popen(
@@ -200,6 +245,7 @@ class ProcRuntime(RuntimeDriver):
 
     async def stop(self, name: str, timeout: int = 10) -> None:
         p = self._get(name)
+        p.manually_stopped = True
         self._refresh(p)
         if p.popen is not None and p.popen.poll() is None:
             self._signal_group(p, signal.SIGTERM)
@@ -257,7 +303,18 @@ class ProcRuntime(RuntimeDriver):
             except OSError:
                 pass
         self._procs.pop(name, None)
-        shutil.rmtree(self._cdir(name), ignore_errors=True)
+        cdir = self._cdir(name)
+        # small trees inline; anything still holding bulk data goes off-loop
+        try:
+            big = any(True for _ in os.scandir(os.path.join(cdir, "rootfs")))
+        except OSError:
+            big = False
+        if big:
+            await asyncio.get_running_loop().run_in_executor(
+                None, lambda: shutil.rmtree(cdir, ignore_errors=True)
+            )
+        else:
+            shutil.rmtree(cdir, ignore_errors=True)
 
     async def inspect(self, name: str) -> Optional[ContainerState]:
         p = self._procs.get(name)
@@ -296,8 +353,13 @@ class ProcRuntime(RuntimeDriver):
         p = self._get(name)
         ref = f"{image}:{tag}" if tag else image
         dest = self._image_dir(ref)
-        shutil.rmtree(dest, ignore_errors=True)
-        shutil.copytree(p.state.upper_dir, dest, symlinks=True)
+        src = p.state.upper_dir
+
+        def _snapshot():
+            shutil.rmtree(dest, ignore_errors=True)
+            shutil.copytree(src, dest, symlinks=True)
+
+        await asyncio.get_running_loop().run_in_executor(None, _snapshot)
         return ref
 
     # --------------------------------------------------------------- volumes
@@ -324,6 +386,9 @@ class ProcRuntime(RuntimeDriver):
         return self.volumes.get(name)
 
     async def close(self) -> None:
+        if self._supervisor is not None:
+            self._supervisor.cancel()
+            self._supervisor = None
         for name in list(self._procs):
             p = self._procs[name]
             if p.popen is not None and p.popen.poll() is None:

@@ -24,6 +24,7 @@ create->running / patch turnaround latency.
 """
 from __future__ import annotations
 
+import asyncio
 import logging
 import os
 import shutil
@@ -464,7 +465,10 @@ class ReplicaSetService:
                 self.ports.restore(self._host_ports(st))
         self._released.discard(vname)
         # wipe preserved layers + merge map entries + version map + store key
-        shutil.rmtree(os.path.join(self.cfg.merges_dir, name), ignore_errors=True)
+        merges_path = os.path.join(self.cfg.merges_dir, name)
+        await asyncio.get_running_loop().run_in_executor(
+            None, lambda: shutil.rmtree(merges_path, ignore_errors=True)
+        )
         self.merges.remove_prefix(name + "-")
         self.versions.remove(name)
         await self.store.delete(Resource.CONTAINERS, name)

@@ -54,6 +54,34 @@ def test_exited_process_detected(tmp_path, run):
     run(main())
 
 
+def test_restart_policy_supervisor(tmp_path, run):
+    import asyncio
+
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        spec = _spec("c-1", cmd=["sh", "-c", "sleep 0.2"])
+        spec.host_config["RestartPolicy"] = {"Name": "unless-stopped"}
+        await rt.create(spec)
+        await rt.start("c-1")
+        first_pid = (await rt.inspect("c-1")).pid
+        # process exits after 0.2 s; supervisor must bring it back
+        restarted = False
+        for _ in range(60):
+            await asyncio.sleep(0.1)
+            st = await rt.inspect("c-1")
+            if st.running and st.pid != first_pid:
+                restarted = True
+                break
+        assert restarted, "supervisor did not restart the exited container"
+        # manual stop must NOT be resurrected
+        await rt.stop("c-1")
+        await asyncio.sleep(0.8)
+        assert not (await rt.inspect("c-1")).running
+        await rt.close()
+
+    run(main())
+
+
 def test_exec_runs_in_rootfs_with_env(tmp_path, run):
     async def main():
         rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
