@@ -199,3 +199,21 @@ def _mount(app: FastAPI, d: Daemon) -> None:
     app.include_router(replicaset_router.make_router(d.replicaset))
     app.include_router(volume_router.make_router(d.volume))
     app.include_router(resource_router.make_router(d.gpu, d.cpu, d.ports))
+
+    from .response import error, success
+    from .codes import Code
+
+    @app.post("/api/v1/admin/compact")
+    async def compact(request: dict | None = None):
+        """MI355X extension: discard state history below a revision and
+        shrink the WAL (memory backend only). Destroys rollback targets
+        below the point — an explicit operator action."""
+        store = d.store
+        if not hasattr(store, "compact"):
+            return error(Code.SERVER_BUSY, detail="state backend does not support compaction")
+        rev = int((request or {}).get("revision", 0))
+        try:
+            result = await store.compact(rev)
+        except Exception as exc:  # noqa: BLE001
+            return error(Code.SERVER_BUSY, detail=str(exc))
+        return success(result)

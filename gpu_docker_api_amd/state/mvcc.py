@@ -68,7 +68,12 @@ class MemoryMVCC:
         self._rev = 1            # etcd: store starts at revision 1
         self._compacted = 0      # highest compacted revision (exclusive floor)
         self._hist: Dict[str, List[_Event]] = {}
-        self.on_event: Optional[Callable[[str, int, Optional[str]], None]] = None
+        # on_event(key, rev, value, version, create_revision) — value None
+        # for tombstones. Version/create_revision ride along so a WAL
+        # rewritten after compaction replays with exact per-key numbering.
+        self.on_event: Optional[
+            Callable[[str, int, Optional[str], int, int], None]
+        ] = None
 
     # ------------------------------------------------------------------ info
     @property
@@ -94,7 +99,7 @@ class MemoryMVCC:
             events.append(_Event(rev, value, version, create_rev))
             self._rev = rev
             if self.on_event:
-                self.on_event(key, rev, value)
+                self.on_event(key, rev, value, version, create_rev)
             return KeyValue(key, value, create_rev, rev, version)
 
     def delete(self, key: str) -> int:
@@ -111,7 +116,7 @@ class MemoryMVCC:
             events.append(_Event(rev, None, 0, 0))
             self._rev = rev
             if self.on_event:
-                self.on_event(key, rev, None)
+                self.on_event(key, rev, None, 0, 0)
             return 1
 
     def delete_prefix(self, prefix: str) -> int:
@@ -129,7 +134,7 @@ class MemoryMVCC:
             for k in sorted(live):
                 self._hist[k].append(_Event(rev, None, 0, 0))
                 if self.on_event:
-                    self.on_event(k, rev, None)
+                    self.on_event(k, rev, None, 0, 0)
             self._rev = rev
             return len(live)
 
@@ -204,14 +209,24 @@ class MemoryMVCC:
             if not events or events[-1].value is None:
                 raise NotExistInStore(key)
             create_rev = events[-1].create_revision
-            out = [
+            lifetime = [e for e in events if e.value is not None and e.create_revision == create_rev]
+            # events above the compaction floor, PLUS the floor itself: the
+            # state at the compaction revision stays readable in etcd
+            # (Get(WithRev=compacted+1) returns it), so the walker would
+            # surface it too.
+            floor = None
+            out_events = []
+            for e in lifetime:
+                if e.mod_revision > self._compacted:
+                    out_events.append(e)
+                else:
+                    floor = e
+            if floor is not None:
+                out_events.insert(0, floor)
+            return [
                 KeyValue(key, e.value, e.create_revision, e.mod_revision, e.version)
-                for e in reversed(events)
-                if e.value is not None
-                and e.create_revision == create_rev
-                and e.mod_revision > self._compacted
+                for e in reversed(out_events)
             ]
-            return out
 
     def get_version(self, key: str, version: int) -> KeyValue:
         """The key-value whose per-key Version equals ``version`` (current
@@ -249,18 +264,39 @@ class MemoryMVCC:
             self._compacted = rev - 1 if rev > 0 else 0
 
     # ---------------------------------------------------------------- replay
-    def replay(self, key: str, rev: int, value: Optional[str]) -> None:
-        """Re-apply a WAL record. Only valid in ascending rev order."""
+    def replay(
+        self,
+        key: str,
+        rev: int,
+        value: Optional[str],
+        version: Optional[int] = None,
+        create_rev: Optional[int] = None,
+    ) -> None:
+        """Re-apply a WAL record. Only valid in ascending rev order.
+        Explicit version/create_rev (post-compaction WALs) take precedence
+        over the computed lifetime numbering."""
         with self._lock:
             events = self._hist.setdefault(key, [])
             last = events[-1] if events else None
             if value is None:
                 events.append(_Event(rev, None, 0, 0))
             else:
-                if last is None or last.value is None:
-                    create_rev, version = rev, 1
-                else:
-                    create_rev, version = last.create_revision, last.version + 1
+                if version is None or create_rev is None:
+                    if last is None or last.value is None:
+                        create_rev, version = rev, 1
+                    else:
+                        create_rev, version = last.create_revision, last.version + 1
                 events.append(_Event(rev, value, version, create_rev))
             if rev > self._rev:
                 self._rev = rev
+
+    def dump_events(self):
+        """All retained events in revision order (WAL rewrite after
+        compaction): (key, rev, value|None, version, create_revision)."""
+        with self._lock:
+            out = []
+            for key, events in self._hist.items():
+                for e in events:
+                    out.append((key, e.mod_revision, e.value, e.version, e.create_revision))
+            out.sort(key=lambda t: t[1])
+            return out

@@ -83,6 +83,15 @@ class MemoryStore(StateStore):
     async def get_version(self, resource: Resource, key: str, version: int) -> KeyValue:
         return self.mvcc.get_version(resource_key(resource, key), version)
 
+    async def compact(self, revision: int = 0) -> dict:
+        """Discard history below ``revision`` (default: everything but the
+        live state) and shrink the WAL. History queries below the point
+        raise RevisionCompacted afterwards — an explicit operator action."""
+        rev = revision or self.mvcc.revision
+        self.mvcc.compact(rev)
+        wal_bytes = self._wal.rewrite() if self._wal else 0
+        return {"compacted_revision": rev, "wal_bytes": wal_bytes}
+
     async def close(self) -> None:
         if self._wal:
             self._wal.close()
