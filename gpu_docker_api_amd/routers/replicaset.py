@@ -12,7 +12,7 @@ from ..models import (
     PatchRequest,
     RollbackRequest,
 )
-from ..models.memory import MEMORY_UNITS, parse_size
+from ..models.memory import parse_size
 from ..services.replicaset import ReplicaSetService
 from .codes import Code
 from .errors import log_error, map_error

@@ -13,7 +13,7 @@ by callers that want them).
 from __future__ import annotations
 
 import grp
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from .base import GpuResolver
 
