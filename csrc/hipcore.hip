@@ -311,6 +311,74 @@ double p2p_bandwidth_gbps(int src_dev, int dst_dev, int mib, int iters) {
   return (double)(n * sizeof(float)) / (ms * 1e6);
 }
 
+// Full pairwise matrix in one call: per-device src/dst buffers allocated
+// ONCE, peer access enabled once, then every ordered pair timed. On an
+// 8-GPU node this is ~56 measurements; per-pair re-allocation would push
+// daemon startup to minutes.
+std::vector<std::vector<double>> p2p_matrix(int mib, int iters) {
+  int n = 0;
+  HIP_CHECK(hipGetDeviceCount(&n));
+  std::vector<std::vector<double>> out(n, std::vector<double>(n, 0.0));
+  if (n == 0) return out;
+  size_t count = (size_t)mib * 1024 * 1024 / sizeof(float);
+
+  std::vector<float*> src(n, nullptr), dst(n, nullptr);
+  for (int i = 0; i < n; ++i) {
+    HIP_CHECK(hipSetDevice(i));
+    HIP_CHECK(hipMalloc(&src[i], count * sizeof(float)));
+    HIP_CHECK(hipMalloc(&dst[i], count * sizeof(float)));
+    HIP_CHECK(hipMemset(src[i], 1, count * sizeof(float)));
+    mfma_warmup(i, 2000);
+    for (int j = 0; j < n; ++j) {
+      if (i == j) continue;
+      int can = 0;
+      HIP_CHECK(hipDeviceCanAccessPeer(&can, i, j));
+      if (can) {
+        hipError_t e = hipDeviceEnablePeerAccess(j, 0);
+        if (e != hipSuccess && e != hipErrorPeerAccessAlreadyEnabled)
+          (void)hipGetLastError();
+        else
+          (void)hipGetLastError();
+      }
+    }
+  }
+  for (int d = 0; d < n; ++d) {
+    for (int s = 0; s < n; ++s) {
+      if (s == d) {
+        double ms = time_kernel_ms(
+            d, [&](hipStream_t st) { launch_copy_f32(src[d], dst[d], count, st); },
+            iters);
+        out[s][d] = (2.0 * count * sizeof(float)) / (ms * 1e6);
+        continue;
+      }
+      int can = 0;
+      HIP_CHECK(hipDeviceCanAccessPeer(&can, d, s));
+      double ms;
+      if (can) {
+        // dst-device kernel pulls from src-device memory over xGMI
+        ms = time_kernel_ms(
+            d, [&](hipStream_t st) { launch_copy_f32(src[s], dst[d], count, st); },
+            iters);
+      } else {
+        ms = time_kernel_ms(
+            d,
+            [&](hipStream_t st) {
+              HIP_CHECK(hipMemcpyPeerAsync(dst[d], d, src[s], s,
+                                           count * sizeof(float), st));
+            },
+            iters);
+      }
+      out[s][d] = (double)(count * sizeof(float)) / (ms * 1e6);
+    }
+  }
+  for (int i = 0; i < n; ++i) {
+    (void)hipSetDevice(i);
+    (void)hipFree(src[i]);
+    (void)hipFree(dst[i]);
+  }
+  return out;
+}
+
 // ---------------------------------------------------------------------------
 // Torch bindings
 // ---------------------------------------------------------------------------
@@ -415,5 +483,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::call_guard<py::gil_scoped_release>());
   m.def("p2p_bandwidth_gbps", &p2p_bandwidth_gbps, py::arg("src_dev"),
         py::arg("dst_dev"), py::arg("mib") = 512, py::arg("iters") = 10,
+        py::call_guard<py::gil_scoped_release>());
+  m.def("p2p_matrix", &p2p_matrix, py::arg("mib") = 256, py::arg("iters") = 5,
         py::call_guard<py::gil_scoped_release>());
 }

@@ -51,7 +51,7 @@ def gpu_available() -> bool:
         return False
 
 
-def run_probe(mib: int = 512, iters: int = 5) -> dict:
+def run_probe(mib: int = 256, iters: int = 5) -> dict:
     """Measure HBM stream + pairwise p2p bandwidth across visible GPUs.
 
     Returns {"gpus": [uuid...], "hbm_gbps": [...], "p2p_gbps": [[...]]} —
@@ -72,16 +72,17 @@ def run_probe(mib: int = 512, iters: int = 5) -> dict:
             uuids = [g.uuid for g in infos]
     except Exception:
         pass
-    hbm = [round(ext.stream_bandwidth_gbps(i, mib, iters), 1) for i in range(n)]
-    p2p = [[0.0] * n for _ in range(n)]
-    for i in range(n):
-        for j in range(n):
-            if i != j:
-                p2p[i][j] = round(ext.p2p_bandwidth_gbps(i, j, mib, iters), 1)
+    # one-shot matrix: per-device buffers allocated once (on an 8-GPU node
+    # per-pair re-allocation would cost minutes of daemon startup)
+    mat = ext.p2p_matrix(mib, iters)
+    hbm = [round(mat[i][i], 1) for i in range(n)]
+    p2p = [
+        [0.0 if i == j else round(mat[i][j], 1) for j in range(n)] for i in range(n)
+    ]
     return {"gpus": uuids, "hbm_gbps": hbm, "p2p_gbps": p2p}
 
 
-async def run_probe_async(mib: int = 512, iters: int = 5) -> dict:
+async def run_probe_async(mib: int = 256, iters: int = 5) -> dict:
     return await asyncio.get_running_loop().run_in_executor(
         None, lambda: run_probe(mib, iters)
     )
