@@ -60,6 +60,17 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
             return error(map_error(exc, Code.CONTAINER_RUN_FAILED))
         return success(data)
 
+    @r.get("")
+    async def list_all():
+        """MI355X extension: list all replicaSets (the reference has no
+        list route; its clients must track names externally)."""
+        try:
+            data = await svc.list_replicasets()
+        except Exception as exc:  # noqa: BLE001
+            log_error("replicaSet.list", exc)
+            return error(map_error(exc, Code.CONTAINER_GET_INFO_FAILED))
+        return success(data)
+
     @r.get("/{name}")
     async def info(name: str):
         try:

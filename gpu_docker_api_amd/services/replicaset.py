@@ -491,6 +491,32 @@ class ReplicaSetService:
         return await self.runtime.commit(vname, req.new_image_name)
 
     # ------------------------------------------------------------------ info
+    async def list_replicasets(self) -> List[Dict]:
+        """All replicaSets with their stored spec + live runtime state
+        (extension: the reference has no list endpoint)."""
+        out = []
+        live = {st.name: st for st in await self.runtime.list(all=True)}
+        for name, version in sorted(self.versions.snapshot().items()):
+            kv = await self.store.get_or_none(Resource.CONTAINERS, name)
+            if kv is None:
+                continue
+            spec = ContainerSpec.deserialize(kv.value)
+            st = live.get(spec.container_name)
+            out.append(
+                {
+                    "name": name,
+                    "containerName": spec.container_name,
+                    "version": version,
+                    "image": spec.image,
+                    "gpuCount": len(spec.gpu_uuids),
+                    "cpuset": spec.cpuset_cpus,
+                    "memory": spec.memory_bytes,
+                    "status": st.status if st else "unknown",
+                    "running": bool(st.running) if st else False,
+                }
+            )
+        return out
+
     async def get_container_info(self, name: str) -> Dict:
         kv = await self.store.get_or_none(Resource.CONTAINERS, name)
         if kv is None:
