@@ -183,6 +183,27 @@ __global__ void mfma_warmup_kernel(float* __restrict__ sink, int iters) {
   if (sink != nullptr && threadIdx.x == 0 && blockIdx.x == 0) sink[0] = acc[0];
 }
 
+}  // namespace  (reopened below — gemm lives in its own TU-style include)
+
+#include "gemm_bf16.hip"
+
+namespace {
+
+// LCG-hash fill: pseudorandom bf16 in [-1, 1) — zero-filled operands
+// overstate GEMM throughput on this chip (guide §5.4 rule 25).
+__global__ void fill_bf16_hash_kernel(__hip_bfloat16* p, size_t n, unsigned seed) {
+  size_t i = blockIdx.x * (size_t)blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    unsigned h = (unsigned)i * 2654435761u + seed;
+    h ^= h >> 15;
+    h *= 2246822519u;
+    h ^= h >> 13;
+    float v = ((h & 0xFFFFFF) / 8388608.0f) - 1.0f;  // [-1, 1)
+    p[i] = __hip_bfloat16(v);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Host-side probe machinery
 // ---------------------------------------------------------------------------
@@ -440,6 +461,54 @@ torch::Tensor mfma_bf16_matmul(torch::Tensor A, torch::Tensor B) {
   return C;
 }
 
+torch::Tensor gemm_bf16_bt(torch::Tensor A, torch::Tensor Bt) {
+  TORCH_CHECK(A.is_cuda() && Bt.is_cuda(), "GPU tensors required");
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+              Bt.scalar_type() == torch::kBFloat16, "bf16 only");
+  A = A.contiguous();
+  Bt = Bt.contiguous();
+  int M = A.size(0), K = A.size(1), N = Bt.size(0);
+  TORCH_CHECK(Bt.size(1) == K, "Bt must be [N][K]");
+  TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 32 == 0,
+              "M,N multiples of 128; K multiple of 32");
+  auto C = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
+  int grid = (M / 128) * (N / 128);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(gemm_bf16::gemm_bf16_tile_kernel, dim3(grid), dim3(256),
+                     0, stream.stream(),
+                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(Bt.data_ptr()),
+                     C.data_ptr<float>(), M, N, K);
+  return C;
+}
+
+// Dense bf16 throughput on one device (the health-check burn-in number).
+double gemm_bf16_tflops(int device, int size, int iters) {
+  HIP_CHECK(hipSetDevice(device));
+  mfma_warmup(device, 20000);
+  size_t n = (size_t)size * size;
+  __hip_bfloat16 *A = nullptr, *Bt = nullptr;
+  float* C = nullptr;
+  HIP_CHECK(hipMalloc(&A, n * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&Bt, n * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&C, n * sizeof(float)));
+  hipLaunchKernelGGL(fill_bf16_hash_kernel, dim3(4096), dim3(256), 0, 0, A, n, 1u);
+  hipLaunchKernelGGL(fill_bf16_hash_kernel, dim3(4096), dim3(256), 0, 0, Bt, n, 7u);
+  HIP_CHECK(hipDeviceSynchronize());
+  int grid = (size / 128) * (size / 128);
+  double ms = time_kernel_ms(
+      device,
+      [&](hipStream_t s) {
+        hipLaunchKernelGGL(gemm_bf16::gemm_bf16_tile_kernel, dim3(grid),
+                           dim3(256), 0, s, A, Bt, C, size, size, size);
+      },
+      iters);
+  (void)hipFree(A);
+  (void)hipFree(Bt);
+  (void)hipFree(C);
+  return 2.0 * size * (double)size * size / (ms * 1e9);
+}
+
 int device_count() {
   int n = 0;
   if (hipGetDeviceCount(&n) != hipSuccess) return 0;
@@ -475,6 +544,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_bf16_matmul", &mfma_bf16_matmul, py::arg("A"), py::arg("B"));
   m.def("mfma_warmup", &mfma_warmup, py::arg("device") = 0,
         py::arg("spins") = 20000);
+  m.def("gemm_bf16_bt", &gemm_bf16_bt, py::arg("A"), py::arg("Bt"));
+  m.def("gemm_bf16_tflops", &gemm_bf16_tflops, py::arg("device") = 0,
+        py::arg("size") = 4096, py::arg("iters") = 10,
+        py::call_guard<py::gil_scoped_release>());
   m.def("stream_bandwidth_gbps", &stream_bandwidth_gbps, py::arg("device") = 0,
         py::arg("mib") = 1024, py::arg("iters") = 10,
         py::call_guard<py::gil_scoped_release>());

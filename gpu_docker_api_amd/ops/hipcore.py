@@ -88,6 +88,40 @@ async def run_probe_async(mib: int = 256, iters: int = 5) -> dict:
     )
 
 
+def validate_gpus(indices: Optional[list] = None, size: int = 4096, iters: int = 5) -> dict:
+    """Per-GPU health burn-in: HBM stream bandwidth + dense bf16 MFMA GEMM
+    throughput (guide's 128^2-tile structure). Run on FREE GPUs before
+    scheduling; a GPU far below its siblings is flagged."""
+    ext = load_ext()
+    n = ext.device_count()
+    idx = list(indices) if indices else list(range(n))
+    report = {"gpus": []}
+    for i in idx:
+        if i < 0 or i >= n:
+            continue
+        hbm = ext.stream_bandwidth_gbps(i, 1024, 5)
+        tflops = ext.gemm_bf16_tflops(i, size, iters)
+        report["gpus"].append(
+            {
+                "index": i,
+                "hbm_gbps": round(hbm, 1),
+                "bf16_tflops": round(tflops, 1),
+            }
+        )
+    vals = [g["bf16_tflops"] for g in report["gpus"]]
+    if vals:
+        top = max(vals)
+        for g in report["gpus"]:
+            g["healthy"] = bool(g["bf16_tflops"] > 0.7 * top and g["hbm_gbps"] > 2000)
+    return report
+
+
+async def validate_gpus_async(indices: Optional[list] = None, size: int = 4096, iters: int = 5) -> dict:
+    return await asyncio.get_running_loop().run_in_executor(
+        None, lambda: validate_gpus(indices, size, iters)
+    )
+
+
 def rccl_smoke_path() -> str:
     root = os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     return os.path.join(root, "csrc", "bin", "rccl_smoke")

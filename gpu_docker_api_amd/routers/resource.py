@@ -22,6 +22,37 @@ def make_router(gpu: GpuScheduler, cpu: CpuScheduler, ports: PortScheduler) -> A
         # MI355X extension: HBM bytes + xGMI adjacency for job sizing
         return success(gpu.get_detail())
 
+    @r.post("/gpus/validate")
+    async def gpus_validate(body: dict | None = None):
+        """MI355X extension: burn-in the node's FREE GPUs (HBM bandwidth +
+        dense bf16 MFMA GEMM) before trusting them with placements. Busy
+        GPUs are skipped unless explicitly listed."""
+        body = body or {}
+        requested = body.get("gpus")
+        status = gpu.get_gpu_status()
+        if requested:
+            idx = [
+                g.index
+                for u in requested
+                if (g := gpu.info_by_uuid(u)) is not None
+            ]
+        else:
+            idx = [
+                g.index for g in gpu.gpus if status.get(g.uuid, 1) == 0
+            ]
+        try:
+            from ..ops import hipcore
+
+            report = await hipcore.validate_gpus_async(
+                idx, int(body.get("size", 4096)), int(body.get("iters", 5))
+            )
+        except Exception as exc:  # noqa: BLE001 — no GPU / extension absent
+            from .codes import Code
+            from .response import error
+
+            return error(Code.SERVER_BUSY, detail=f"validation unavailable: {exc}")
+        return success(report)
+
     @r.get("/cpus")
     async def cpus():
         return success(cpu.get_cpu_status())

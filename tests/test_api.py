@@ -169,6 +169,16 @@ def test_saturation_scenario(client):
     assert client.post("/api/v1/replicaSet", json=body).json()["code"] == 200
 
 
+def test_gpu_validate_endpoint_graceful_without_gpu(client):
+    r = client.post("/api/v1/resources/gpus/validate", json={})
+    body = r.json()
+    # CPU box: either an empty report (extension loads, 0 devices) or a
+    # clean 500-coded error — never a crash
+    assert body["code"] in (200, 500)
+    if body["code"] == 200:
+        assert body["data"]["gpus"] == []
+
+
 def test_apikey_auth(tmp_path):
     cfg = make_config(tmp_path, apikey="secret-token")
     app = build_app(cfg)
