@@ -65,9 +65,23 @@ extern "C" __global__ __launch_bounds__(THREADS) void gemm_bf16_tile_kernel(
     int M, int N, int K) {
   __shared__ __hip_bfloat16 lds[2][2][TILE_HALFS];  // [dbuf][A/B][tile]
 
+  // XCD-aware remap: the dispatcher places block b on XCD b%8, so renumber
+  // blocks XCD-major to keep neighboring tiles on one XCD's L2. Bijective
+  // form (guide §5: the naive remap is non-bijective when nwg%8 != 0).
+  // Pays ~+10% when HBM-bound (large N); gated off for small grids where
+  // the working set is L3-resident and the remap costs ~2%.
+  int wgid = blockIdx.x;
+  const int nwg = gridDim.x;
+  if (nwg >= 2048) {
+    const int xcd = wgid & 7;
+    const int pos = wgid >> 3;
+    const int q = nwg >> 3;
+    const int r = nwg & 7;
+    wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
   const int tiles_n = N / BN;
-  const int bm = (blockIdx.x / tiles_n) * BM;
-  const int bn = (blockIdx.x % tiles_n) * BN;
+  const int bm = (wgid / tiles_n) * BM;
+  const int bn = (wgid % tiles_n) * BN;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int wm = (wave / WAVES_N) * (BM / WAVES_M);  // 0 or 64

@@ -220,6 +220,10 @@ class ReplicaSetService:
         changed = False
 
         async with Saga("patch") as saga:
+            if vname_old in self._released:
+                # stopped container: its resources were released at stop —
+                # re-acquire them first or the replacement double-books GPUs
+                self._reacquire_released(saga, spec, vname_old)
             if req.gpu_patch is not None:
                 changed |= self._patch_gpu(saga, spec, req.gpu_patch.gpu_count)
             if req.cpu_patch is not None:
@@ -238,6 +242,18 @@ class ReplicaSetService:
             saga.commit()
         phases = timer.finish()
         return {"containerName": vname_new, "id": cid, "phases": phases}
+
+    def _reacquire_released(self, saga: Saga, spec: ContainerSpec, vname: str) -> None:
+        """Re-acquire the exact resources a stop released (raises when any
+        is now taken by another replicaSet)."""
+        if spec.gpu_uuids:
+            self.gpu.apply_specific(spec.gpu_uuids)
+            saga.push("gpu-reacquired", lambda: self.gpu.restore(spec.gpu_uuids))
+        if spec.cpuset_cpus:
+            self.cpu.apply_specific(spec.cpuset_cpus)
+            saga.push("cpu-reacquired", lambda: self.cpu.restore(spec.cpuset_cpus))
+        self._released.discard(vname)
+        saga.push("mark-released", lambda: self._released.add(vname))
 
     def _patch_gpu(self, saga: Saga, spec: ContainerSpec, new_count: int) -> bool:
         old_uuids = spec.gpu_uuids
@@ -365,6 +381,8 @@ class ReplicaSetService:
         vname_old = versioned(name, cur_version)
 
         async with Saga("rollback") as saga:
+            if vname_old in self._released:
+                self._reacquire_released(saga, live, vname_old)
             # re-resolve GPU/CPU against live allocation state
             self._patch_gpu(saga, live, len(target.gpu_uuids))
             self._patch_cpu(saga, live, len([c for c in target.cpuset_cpus.split(",") if c]))

@@ -222,6 +222,36 @@ def test_stop_releases_and_startup_reacquires(tmp_path, run):
     run(main())
 
 
+def test_patch_of_stopped_container_reacquires_resources(tmp_path, run):
+    """A stopped replicaSet released its GPUs; patching it must re-acquire
+    them (or fail) — otherwise the replacement double-books."""
+
+    async def main():
+        d = await make_daemon(tmp_path)
+        await d.replicaset.run_gpu_container(_run_req(name="a", gpus=2))
+        await d.replicaset.stop_container("a")
+        assert sum(d.gpu.get_gpu_status().values()) == 0
+        # another tenant takes 7 of 8 GPUs
+        await d.replicaset.run_gpu_container(_run_req(name="b", gpus=7))
+        # patch a: its old 2 GPUs can't all be re-acquired -> must fail clean
+        with pytest.raises(GpuNotEnough):
+            await d.replicaset.patch_container(
+                "a", PatchRequest(gpu_patch=GpuPatch(gpu_count=2))
+            )
+        assert sum(d.gpu.get_gpu_status().values()) == 7  # unchanged
+        await d.replicaset.delete_container("b")
+        # now it works: re-acquire + rescale
+        out = await d.replicaset.patch_container(
+            "a", PatchRequest(gpu_patch=GpuPatch(gpu_count=3))
+        )
+        st = await d.runtime.inspect(out["containerName"])
+        assert len(st.gpu_uuids) == 3
+        assert sum(d.gpu.get_gpu_status().values()) == 3
+        await d.stop()
+
+    run(main())
+
+
 def test_pause_keeps_resources(tmp_path, run):
     async def main():
         d = await make_daemon(tmp_path)
