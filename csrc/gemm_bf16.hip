@@ -62,7 +62,7 @@ extern "C" __global__ __launch_bounds__(THREADS) void gemm_bf16_tile_kernel(
     const __hip_bfloat16* __restrict__ A,   // [M][K]
     const __hip_bfloat16* __restrict__ Bt,  // [N][K] (B transposed)
     float* __restrict__ C,                  // [M][N]
-    int M, int N, int K) {
+    int M, int N, int K, int use_swizzle) {
   __shared__ __hip_bfloat16 lds[2][2][TILE_HALFS];  // [dbuf][A/B][tile]
 
   // XCD-aware remap: the dispatcher places block b on XCD b%8, so renumber
@@ -72,7 +72,7 @@ extern "C" __global__ __launch_bounds__(THREADS) void gemm_bf16_tile_kernel(
   // the working set is L3-resident and the remap costs ~2%.
   int wgid = blockIdx.x;
   const int nwg = gridDim.x;
-  if (nwg >= 2048) {
+  if (use_swizzle) {
     const int xcd = wgid & 7;
     const int pos = wgid >> 3;
     const int q = nwg >> 3;

@@ -478,12 +478,12 @@ torch::Tensor gemm_bf16_bt(torch::Tensor A, torch::Tensor Bt) {
                      0, stream.stream(),
                      reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
                      reinterpret_cast<const __hip_bfloat16*>(Bt.data_ptr()),
-                     C.data_ptr<float>(), M, N, K);
+                     C.data_ptr<float>(), M, N, K, /*swizzle=*/0);
   return C;
 }
 
 // Dense bf16 throughput on one device (the health-check burn-in number).
-double gemm_bf16_tflops(int device, int size, int iters) {
+double gemm_bf16_tflops(int device, int size, int iters, int swizzle) {
   HIP_CHECK(hipSetDevice(device));
   mfma_warmup(device, 20000);
   size_t n = (size_t)size * size;
@@ -500,7 +500,7 @@ double gemm_bf16_tflops(int device, int size, int iters) {
       device,
       [&](hipStream_t s) {
         hipLaunchKernelGGL(gemm_bf16::gemm_bf16_tile_kernel, dim3(grid),
-                           dim3(256), 0, s, A, Bt, C, size, size, size);
+                           dim3(256), 0, s, A, Bt, C, size, size, size, swizzle);
       },
       iters);
   (void)hipFree(A);
@@ -546,7 +546,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("spins") = 20000);
   m.def("gemm_bf16_bt", &gemm_bf16_bt, py::arg("A"), py::arg("Bt"));
   m.def("gemm_bf16_tflops", &gemm_bf16_tflops, py::arg("device") = 0,
-        py::arg("size") = 4096, py::arg("iters") = 10,
+        py::arg("size") = 4096, py::arg("iters") = 10, py::arg("swizzle") = 0,
         py::call_guard<py::gil_scoped_release>());
   m.def("stream_bandwidth_gbps", &stream_bandwidth_gbps, py::arg("device") = 0,
         py::arg("mib") = 1024, py::arg("iters") = 10,

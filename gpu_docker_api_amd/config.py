@@ -36,6 +36,11 @@ class Config:
     # copy engine: auto | iouring | tar | python
     copy_engine: str = "auto"
 
+    # how many replaced versions' writable layers to keep under merges/
+    # per replicaSet (0 = unlimited). Rollback --restore-data needs the
+    # target version's layer to still be retained.
+    keep_merge_layers: int = 5
+
     # native probes at startup (GPU boxes only)
     run_xgmi_probe: bool = False
     run_rccl_smoke: bool = False

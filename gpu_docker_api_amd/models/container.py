@@ -68,7 +68,10 @@ class PatchRequest(BaseModel):
 
 
 class RollbackRequest(BaseModel):
+    model_config = ConfigDict(populate_by_name=True)
     version: int = 0
+    # extension: also restore the preserved writable layer of that version
+    restore_data: bool = Field(False, alias="restoreData")
 
 
 class ContainerExecute(BaseModel):

@@ -148,7 +148,7 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
         if req.version < 0:
             return error(Code.CONTAINER_VERSION_GE_ZERO)
         try:
-            data = await svc.rollback_container(name, req.version)
+            data = await svc.rollback_container(name, req.version, req.restore_data)
         except Exception as exc:  # noqa: BLE001
             log_error("replicaSet.rollback", exc)
             return error(map_error(exc, Code.CONTAINER_ROLLBACK_FAILED))

@@ -303,21 +303,23 @@ class ReplicaSetService:
         spec: ContainerSpec,
         vname_old: str,
         timer: PhaseTimer,
+        migrate_src: Optional[str] = None,
     ) -> Tuple[str, str]:
         """The rolling-replace tail shared by patch/rollback/restart
         (reference replicaset.go:318-359): create new version, migrate the
-        writable layer, start, preserve old layer, delete old."""
+        writable layer (or ``migrate_src`` — rollback --restore-data uses a
+        preserved historical layer), start, preserve old layer, delete old."""
         old_state = await self.runtime.inspect(vname_old)
         cid, vname_new = await self._run_container(saga, name, spec, only_create=True, timer=timer)
         new_state = await self.runtime.inspect(vname_new)
+        src = migrate_src or (old_state.upper_dir if old_state is not None else "")
         if (
-            old_state is not None
-            and new_state is not None
-            and old_state.upper_dir
+            new_state is not None
+            and src
             and new_state.upper_dir
-            and os.path.isdir(old_state.upper_dir)
+            and os.path.isdir(src)
         ):
-            await self.copy.copy_dir(old_state.upper_dir, new_state.upper_dir)
+            await self.copy.copy_dir(src, new_state.upper_dir)
         timer.mark("copy")
         await self.runtime.start(vname_new)
         timer.mark("start")
@@ -340,6 +342,7 @@ class ReplicaSetService:
                     os.makedirs(merge_path, exist_ok=True)
                     await self.copy.copy_dir(old_state.upper_dir, merge_path)
                 self.merges.set(vname_old, merge_path)
+                self._prune_merge_layers(name)
             except Exception:
                 log.exception("preserving %s layer failed", vname_old)
         timer.mark("preserve")
@@ -350,6 +353,24 @@ class ReplicaSetService:
         self._released.discard(vname_old)
         timer.mark("delete_old")
         return vname_new, cid
+
+    def _prune_merge_layers(self, name: str) -> None:
+        """Retain only the newest cfg.keep_merge_layers preserved layers."""
+        keep = self.cfg.keep_merge_layers
+        if keep <= 0:
+            return
+        prefix = name + "-"
+        entries = []
+        for vname, path in self.merges.snapshot().items():
+            if vname.startswith(prefix):
+                try:
+                    entries.append((int(vname[len(prefix):]), vname, path))
+                except ValueError:
+                    continue
+        entries.sort(reverse=True)
+        for _v, vname, path in entries[keep:]:
+            self.merges.remove(vname)
+            shutil.rmtree(path, ignore_errors=True)
 
     @staticmethod
     def _host_ports(state: ContainerState) -> List[int]:
@@ -363,12 +384,19 @@ class ReplicaSetService:
         return [p for p in out if p]
 
     # --------------------------------------------------------------- rollback
-    async def rollback_container(self, name: str, target_version: int) -> Dict:
+    async def rollback_container(
+        self, name: str, target_version: int, restore_data: bool = False
+    ) -> Dict:
         """PATCH /{name}/rollback (reference RollbackContainer,
         replicaset.go:365-446): restore the spec of history version N, with
         resources re-resolved against live state. Note: the reference's
         memory restore divides by 1024^2 and labels it GB — a 1024x
-        inflation (replicaset.go:408) — we restore exact bytes."""
+        inflation (replicaset.go:408) — we restore exact bytes.
+
+        ``restore_data`` (extension): also restore that version's preserved
+        writable layer from merges/ instead of migrating the current data —
+        possible here because preservation is real (the reference's is a
+        commented-out no-op, replicaset.go:688-698)."""
         timer = PhaseTimer("replicaset.rollback")
         cur_version = self.versions.get(name)
         if cur_version is None:
@@ -392,7 +420,18 @@ class ReplicaSetService:
             new_spec.cpuset_cpus = live.cpuset_cpus
             new_spec.memory_bytes = target.memory_bytes
             timer.mark("schedule")
-            vname_new, cid = await self._replace(saga, name, new_spec, vname_old, timer)
+            migrate_src = None
+            if restore_data:
+                preserved = self.merges.get(versioned(name, target_version))
+                if not preserved or not os.path.isdir(preserved):
+                    raise NoRollbackRequired(
+                        f"{name}: no preserved layer for version {target_version} "
+                        f"(retention keep_merge_layers={self.cfg.keep_merge_layers})"
+                    )
+                migrate_src = preserved
+            vname_new, cid = await self._replace(
+                saga, name, new_spec, vname_old, timer, migrate_src=migrate_src
+            )
             saga.commit()
         phases = timer.finish()
         return {"containerName": vname_new, "id": cid, "phases": phases}

@@ -199,6 +199,48 @@ def test_rollback_restores_old_spec(tmp_path, run):
     run(main())
 
 
+def test_rollback_with_data_restore(tmp_path, run):
+    async def main():
+        d = await make_daemon(tmp_path)
+        await d.replicaset.run_gpu_container(_run_req(gpus=0, cpus=0))
+        st1 = await d.runtime.inspect("demo-1")
+        with open(os.path.join(st1.upper_dir, "gen.txt"), "w") as f:
+            f.write("generation-1")
+        await d.replicaset.patch_container("demo", PatchRequest())  # -> demo-2
+        st2 = await d.runtime.inspect("demo-2")
+        with open(os.path.join(st2.upper_dir, "gen.txt"), "w") as f:
+            f.write("generation-2")
+        # plain rollback keeps CURRENT data (reference behavior)
+        out = await d.replicaset.rollback_container("demo", 1)
+        st3 = await d.runtime.inspect(out["containerName"])
+        assert open(os.path.join(st3.upper_dir, "gen.txt")).read() == "generation-2"
+        # rollback with restore_data brings back version 1's layer
+        out = await d.replicaset.rollback_container("demo", 1, restore_data=True)
+        st4 = await d.runtime.inspect(out["containerName"])
+        assert open(os.path.join(st4.upper_dir, "gen.txt")).read() == "generation-1"
+        await d.stop()
+
+    run(main())
+
+
+def test_merge_layer_retention_pruning(tmp_path, run):
+    async def main():
+        d = await make_daemon(tmp_path)
+        d.replicaset.cfg.keep_merge_layers = 2
+        await d.replicaset.run_gpu_container(_run_req(gpus=0, cpus=0))
+        for _ in range(5):
+            await d.replicaset.patch_container("demo", PatchRequest())
+        snap = d.replicaset.merges.snapshot()
+        kept = sorted(k for k in snap if k.startswith("demo-"))
+        assert kept == ["demo-4", "demo-5"]
+        # pruned layer dirs are gone from disk
+        assert not os.path.exists(os.path.join(d.cfg.merges_dir, "demo", "demo-1"))
+        assert os.path.exists(snap["demo-5"])
+        await d.stop()
+
+    run(main())
+
+
 def test_stop_releases_and_startup_reacquires(tmp_path, run):
     async def main():
         d = await make_daemon(tmp_path)
