@@ -27,15 +27,11 @@ using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 
 constexpr int BM = 128;
 constexpr int BN = 128;
-constexpr int BK = 32;
 constexpr int WAVES_M = 2;
 constexpr int WAVES_N = 2;
 constexpr int THREADS = WAVES_M * WAVES_N * 64;  // 256
 constexpr int FRAGS_M = BM / WAVES_M / 16;       // 4
 constexpr int FRAGS_N = BN / WAVES_N / 16;       // 4
-
-// one staged tile: 128 rows x 32 bf16 = 8 KiB; A and B^T use the same shape
-constexpr int TILE_HALFS = BM * BK;  // bf16 elements
 
 __device__ inline void glds16(const __hip_bfloat16* gsrc, __hip_bfloat16* lds) {
   __builtin_amdgcn_global_load_lds(
@@ -44,26 +40,30 @@ __device__ inline void glds16(const __hip_bfloat16* gsrc, __hip_bfloat16* lds) {
 }
 
 // Stage a [128][BK] tile from row-major [rows][ld] global memory into linear
-// LDS. 256 threads x 16 B = 4 KiB per pass; 2 passes per 8 KiB tile. The
-// LDS image is lane-linear by construction (glds writes base + lane*16).
+// LDS. 256 threads x 16 B per glds pass; BM*BK*2B / 4 KiB passes per tile.
+// The LDS image is lane-linear by construction (glds writes base + lane*16).
+template <int BK>
 __device__ inline void stage_tile(const __hip_bfloat16* g, int ld,
                                   __hip_bfloat16* lds) {
   const int t = threadIdx.x;
+  constexpr int kSlotsPerRow = BK * 2 / 16;  // 16B slots per row
+  constexpr int kPasses = BM * BK * 2 / (THREADS * 16);
 #pragma unroll
-  for (int pass = 0; pass < 2; ++pass) {
-    int slot = pass * THREADS + t;     // 0..511, 16B each
-    int row = slot >> 2;               // BK*2B / 16B = 4 slots per row
-    int kchunk = (slot & 3) * 8;       // 8 bf16 per 16B slot
+  for (int pass = 0; pass < kPasses; ++pass) {
+    int slot = pass * THREADS + t;
+    int row = slot / kSlotsPerRow;
+    int kchunk = (slot % kSlotsPerRow) * 8;  // 8 bf16 per 16B slot
     glds16(g + row * ld + kchunk, lds + row * BK + kchunk);
   }
 }
 
-extern "C" __global__ __launch_bounds__(THREADS) void gemm_bf16_tile_kernel(
+template <int BK>
+__global__ __launch_bounds__(THREADS) void gemm_bf16_tile_kernel(
     const __hip_bfloat16* __restrict__ A,   // [M][K]
     const __hip_bfloat16* __restrict__ Bt,  // [N][K] (B transposed)
     float* __restrict__ C,                  // [M][N]
     int M, int N, int K, int use_swizzle) {
-  __shared__ __hip_bfloat16 lds[2][2][TILE_HALFS];  // [dbuf][A/B][tile]
+  __shared__ __hip_bfloat16 lds[2][2][BM * BK];  // [dbuf][A/B][tile]
 
   // XCD-aware remap: the dispatcher places block b on XCD b%8, so renumber
   // blocks XCD-major to keep neighboring tiles on one XCD's L2. Bijective
@@ -92,8 +92,8 @@ extern "C" __global__ __launch_bounds__(THREADS) void gemm_bf16_tile_kernel(
 
   f32x4 acc[FRAGS_M][FRAGS_N] = {};
 
-  stage_tile(A + bm * K, K, &lds[0][0][0]);
-  stage_tile(Bt + bn * K, K, &lds[0][1][0]);
+  stage_tile<BK>(A + bm * K, K, &lds[0][0][0]);
+  stage_tile<BK>(Bt + bn * K, K, &lds[0][1][0]);
 
   int buf = 0;
   for (int k0 = 0; k0 < K; k0 += BK) {
@@ -105,30 +105,33 @@ extern "C" __global__ __launch_bounds__(THREADS) void gemm_bf16_tile_kernel(
     // prefetch the NEXT K-tile after the barrier: it overlaps the MFMA
     // phase below and is drained at the next iteration's barrier.
     if (k0 + BK < K) {
-      stage_tile(A + bm * K + (k0 + BK), K, &lds[buf ^ 1][0][0]);
-      stage_tile(Bt + bn * K + (k0 + BK), K, &lds[buf ^ 1][1][0]);
+      stage_tile<BK>(A + bm * K + (k0 + BK), K, &lds[buf ^ 1][0][0]);
+      stage_tile<BK>(Bt + bn * K + (k0 + BK), K, &lds[buf ^ 1][1][0]);
     }
 
     const __hip_bfloat16* As = &lds[buf][0][0];
     const __hip_bfloat16* Bs = &lds[buf][1][0];
-    bf16x8 afrag[FRAGS_M];
-    bf16x8 bfrag[FRAGS_N];
 #pragma unroll
-    for (int fm = 0; fm < FRAGS_M; ++fm) {
-      afrag[fm] = *reinterpret_cast<const bf16x8*>(
-          As + (wm + fm * 16 + frow) * BK + fk);
-    }
+    for (int kk = 0; kk < BK / 32; ++kk) {
+      bf16x8 afrag[FRAGS_M];
+      bf16x8 bfrag[FRAGS_N];
 #pragma unroll
-    for (int fn = 0; fn < FRAGS_N; ++fn) {
-      bfrag[fn] = *reinterpret_cast<const bf16x8*>(
-          Bs + (wn + fn * 16 + frow) * BK + fk);
-    }
-#pragma unroll
-    for (int fm = 0; fm < FRAGS_M; ++fm) {
+      for (int fm = 0; fm < FRAGS_M; ++fm) {
+        afrag[fm] = *reinterpret_cast<const bf16x8*>(
+            As + (wm + fm * 16 + frow) * BK + kk * 32 + fk);
+      }
 #pragma unroll
       for (int fn = 0; fn < FRAGS_N; ++fn) {
-        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+        bfrag[fn] = *reinterpret_cast<const bf16x8*>(
+            Bs + (wn + fn * 16 + frow) * BK + kk * 32 + fk);
+      }
+#pragma unroll
+      for (int fm = 0; fm < FRAGS_M; ++fm) {
+#pragma unroll
+        for (int fn = 0; fn < FRAGS_N; ++fn) {
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[fm], bfrag[fn], acc[fm][fn], 0, 0, 0);
+        }
       }
     }
     buf ^= 1;

@@ -474,16 +474,24 @@ torch::Tensor gemm_bf16_bt(torch::Tensor A, torch::Tensor Bt) {
   auto C = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
   int grid = (M / 128) * (N / 128);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(gemm_bf16::gemm_bf16_tile_kernel, dim3(grid), dim3(256),
-                     0, stream.stream(),
-                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
-                     reinterpret_cast<const __hip_bfloat16*>(Bt.data_ptr()),
-                     C.data_ptr<float>(), M, N, K, /*swizzle=*/0);
+  if (K % 64 == 0) {
+    hipLaunchKernelGGL(gemm_bf16::gemm_bf16_tile_kernel<64>, dim3(grid),
+                       dim3(256), 0, stream.stream(),
+                       reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K, /*swizzle=*/0);
+  } else {
+    hipLaunchKernelGGL(gemm_bf16::gemm_bf16_tile_kernel<32>, dim3(grid),
+                       dim3(256), 0, stream.stream(),
+                       reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K, /*swizzle=*/0);
+  }
   return C;
 }
 
 // Dense bf16 throughput on one device (the health-check burn-in number).
-double gemm_bf16_tflops(int device, int size, int iters, int swizzle) {
+double gemm_bf16_tflops(int device, int size, int iters, int swizzle, int bk) {
   HIP_CHECK(hipSetDevice(device));
   mfma_warmup(device, 20000);
   size_t n = (size_t)size * size;
@@ -499,8 +507,13 @@ double gemm_bf16_tflops(int device, int size, int iters, int swizzle) {
   double ms = time_kernel_ms(
       device,
       [&](hipStream_t s) {
-        hipLaunchKernelGGL(gemm_bf16::gemm_bf16_tile_kernel, dim3(grid),
-                           dim3(256), 0, s, A, Bt, C, size, size, size, swizzle);
+        if (bk == 32) {
+          hipLaunchKernelGGL(gemm_bf16::gemm_bf16_tile_kernel<32>, dim3(grid),
+                             dim3(256), 0, s, A, Bt, C, size, size, size, swizzle);
+        } else {
+          hipLaunchKernelGGL(gemm_bf16::gemm_bf16_tile_kernel<64>, dim3(grid),
+                             dim3(256), 0, s, A, Bt, C, size, size, size, swizzle);
+        }
       },
       iters);
   (void)hipFree(A);
@@ -547,7 +560,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16_bt", &gemm_bf16_bt, py::arg("A"), py::arg("Bt"));
   m.def("gemm_bf16_tflops", &gemm_bf16_tflops, py::arg("device") = 0,
         py::arg("size") = 4096, py::arg("iters") = 10, py::arg("swizzle") = 0,
-        py::call_guard<py::gil_scoped_release>());
+        py::arg("bk") = 64, py::call_guard<py::gil_scoped_release>());
   m.def("stream_bandwidth_gbps", &stream_bandwidth_gbps, py::arg("device") = 0,
         py::arg("mib") = 1024, py::arg("iters") = 10,
         py::call_guard<py::gil_scoped_release>());
