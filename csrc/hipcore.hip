@@ -186,6 +186,7 @@ __global__ void mfma_warmup_kernel(float* __restrict__ sink, int iters) {
 }  // namespace  (reopened below — gemm lives in its own TU-style include)
 
 #include "gemm_bf16.hip"
+#include "gemm_bf16_8phase.hip"
 
 namespace {
 
@@ -522,6 +523,53 @@ double gemm_bf16_tflops(int device, int size, int iters, int swizzle, int bk) {
   return 2.0 * size * (double)size * size / (ms * 1e9);
 }
 
+torch::Tensor gemm_bf16_8ph_bt(torch::Tensor A, torch::Tensor Bt) {
+  TORCH_CHECK(A.is_cuda() && Bt.is_cuda(), "GPU tensors required");
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+              Bt.scalar_type() == torch::kBFloat16, "bf16 only");
+  A = A.contiguous();
+  Bt = Bt.contiguous();
+  int M = A.size(0), K = A.size(1), N = Bt.size(0);
+  TORCH_CHECK(Bt.size(1) == K, "Bt must be [N][K]");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0 && K >= 192,
+              "M,N multiples of 256; K multiple of 64, >= 192");
+  auto C = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
+  int grid = (M / 256) * (N / 256);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(gemm_bf16_8ph::gemm_bf16_8phase_kernel, dim3(grid),
+                     dim3(512), 0, stream.stream(),
+                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(Bt.data_ptr()),
+                     C.data_ptr<float>(), M, N, K);
+  return C;
+}
+
+double gemm_bf16_8ph_tflops(int device, int size, int iters) {
+  HIP_CHECK(hipSetDevice(device));
+  mfma_warmup(device, 20000);
+  size_t n = (size_t)size * size;
+  __hip_bfloat16 *A = nullptr, *Bt = nullptr;
+  float* C = nullptr;
+  HIP_CHECK(hipMalloc(&A, n * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&Bt, n * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&C, n * sizeof(float)));
+  hipLaunchKernelGGL(fill_bf16_hash_kernel, dim3(4096), dim3(256), 0, 0, A, n, 1u);
+  hipLaunchKernelGGL(fill_bf16_hash_kernel, dim3(4096), dim3(256), 0, 0, Bt, n, 7u);
+  HIP_CHECK(hipDeviceSynchronize());
+  int grid = (size / 256) * (size / 256);
+  double ms = time_kernel_ms(
+      device,
+      [&](hipStream_t s) {
+        hipLaunchKernelGGL(gemm_bf16_8ph::gemm_bf16_8phase_kernel, dim3(grid),
+                           dim3(512), 0, s, A, Bt, C, size, size, size);
+      },
+      iters);
+  (void)hipFree(A);
+  (void)hipFree(Bt);
+  (void)hipFree(C);
+  return 2.0 * size * (double)size * size / (ms * 1e9);
+}
+
 int device_count() {
   int n = 0;
   if (hipGetDeviceCount(&n) != hipSuccess) return 0;
@@ -558,6 +606,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_warmup", &mfma_warmup, py::arg("device") = 0,
         py::arg("spins") = 20000);
   m.def("gemm_bf16_bt", &gemm_bf16_bt, py::arg("A"), py::arg("Bt"));
+  m.def("gemm_bf16_8ph", &gemm_bf16_8ph_bt, py::arg("A"), py::arg("Bt"));
+  m.def("gemm_bf16_8ph_tflops", &gemm_bf16_8ph_tflops, py::arg("device") = 0,
+        py::arg("size") = 4096, py::arg("iters") = 10,
+        py::call_guard<py::gil_scoped_release>());
   m.def("gemm_bf16_tflops", &gemm_bf16_tflops, py::arg("device") = 0,
         py::arg("size") = 4096, py::arg("iters") = 10, py::arg("swizzle") = 0,
         py::arg("bk") = 64, py::call_guard<py::gil_scoped_release>());
