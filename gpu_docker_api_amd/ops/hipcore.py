@@ -100,7 +100,12 @@ def validate_gpus(indices: Optional[list] = None, size: int = 4096, iters: int =
         if i < 0 or i >= n:
             continue
         hbm = ext.stream_bandwidth_gbps(i, 1024, 5)
-        tflops = ext.gemm_bf16_tflops(i, size, iters)
+        # prefer the 8-phase 256^2 pipelined structure (~1000 TF) when the
+        # shape allows; fall back to the 128^2 double-buffered kernel
+        if size % 256 == 0 and size >= 192:
+            tflops = ext.gemm_bf16_8ph_tflops(i, size, iters)
+        else:
+            tflops = ext.gemm_bf16_tflops(i, size, iters)
         report["gpus"].append(
             {
                 "index": i,
