@@ -181,11 +181,11 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
     wait_lgkm0_fence();
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
+    for (int kk = 0; kk < 2; ++kk)   // kk OUTER: 8 independent MFMAs between
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
+      for (int fm = 0; fm < 4; ++fm)  // accumulator reuse (dependent-latency
 #pragma unroll
-        for (int kk = 0; kk < 2; ++kk)
+        for (int fn = 0; fn < 2; ++fn)  // hiding, guide §3 MFMA u-bench)
           acc[0][fm][0][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[fm][kk], bfrag[0][fn][kk], acc[0][fm][0][fn], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
@@ -203,11 +203,11 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
     wait_lgkm0_fence();
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
+    for (int kk = 0; kk < 2; ++kk)   // kk OUTER: 8 independent MFMAs between
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
+      for (int fm = 0; fm < 4; ++fm)  // accumulator reuse (dependent-latency
 #pragma unroll
-        for (int kk = 0; kk < 2; ++kk)
+        for (int fn = 0; fn < 2; ++fn)  // hiding, guide §3 MFMA u-bench)
           acc[0][fm][1][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[fm][kk], bfrag[1][fn][kk], acc[0][fm][1][fn], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
@@ -225,11 +225,11 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
     wait_lgkm0_fence();
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
+    for (int kk = 0; kk < 2; ++kk)   // kk OUTER: 8 independent MFMAs between
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
+      for (int fm = 0; fm < 4; ++fm)  // accumulator reuse (dependent-latency
 #pragma unroll
-        for (int kk = 0; kk < 2; ++kk)
+        for (int fn = 0; fn < 2; ++fn)  // hiding, guide §3 MFMA u-bench)
           acc[1][fm][1][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[fm][kk], bfrag[1][fn][kk], acc[1][fm][1][fn], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
@@ -241,11 +241,11 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
     wait_lgkm0_fence();
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
+    for (int kk = 0; kk < 2; ++kk)   // kk OUTER: 8 independent MFMAs between
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
+      for (int fm = 0; fm < 4; ++fm)  // accumulator reuse (dependent-latency
 #pragma unroll
-        for (int kk = 0; kk < 2; ++kk)
+        for (int fn = 0; fn < 2; ++fn)  // hiding, guide §3 MFMA u-bench)
           acc[1][fm][0][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[fm][kk], bfrag[0][fn][kk], acc[1][fm][0][fn], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
