@@ -37,8 +37,12 @@ def _run(cmd: list[str]) -> None:
 
 def build_hipcore(force: bool = False) -> str:
     src = os.path.join(CSRC, "hipcore.hip")
+    includes = [
+        os.path.join(CSRC, "gemm_bf16.hip"),
+        os.path.join(CSRC, "gemm_bf16_8phase.hip"),
+    ]
     out = os.path.join(OPS, "_hipcore.so")
-    if not force and _newer(out, src):
+    if not force and _newer(out, src, *[p for p in includes if os.path.exists(p)]):
         return out
     import torch
     from torch.utils import cpp_extension
