@@ -53,11 +53,14 @@ __device__ inline void glds16(const __hip_bfloat16* gsrc, __hip_bfloat16* lds) {
       (__attribute__((address_space(3))) unsigned int*)lds, 16, 0, 0);
 }
 
-// st_16x32 swizzle: byte ^= ((byte>>9)&1)<<5 over a [row][64 bf16] image
-// == element k-base ^= ((row>>2)&1)<<4. Involution; applied identically on
-// the glds source (below) and the ds_read address (read_frag).
+// LDS XOR swizzle (guide T2 recipe): element k-chunk ^= row&7 (i.e.
+// byte_off ^= (row&7)<<4) — spreads a 16-lane ds_read_b128 group over 8
+// 16-B slots; measured on this kernel: the weaker single-bit st_16x32 form
+// left 12.6M bank conflicts per 4096^3 dispatch (PMC SQ_LDS_BANK_CONFLICT).
+// Involution; applied identically on the glds source and the ds_read
+// address (both-sides-or-neither, rule 21).
 __device__ inline int swz_k(int row, int k) {
-  return k ^ (((row >> 2) & 1) << 4);
+  return k ^ ((row & 7) << 3);
 }
 
 // Stage one [128][64] half-tile: 1024 16-B slots, 512 threads x 2 passes.
