@@ -53,34 +53,44 @@ __device__ inline void glds16(const __hip_bfloat16* gsrc, __hip_bfloat16* lds) {
       (__attribute__((address_space(3))) unsigned int*)lds, 16, 0, 0);
 }
 
-// LDS XOR swizzle (guide T2 recipe): element k-chunk ^= row&7 (i.e.
-// byte_off ^= (row&7)<<4) — spreads a 16-lane ds_read_b128 group over 8
-// 16-B slots; measured on this kernel: the weaker single-bit st_16x32 form
-// left 12.6M bank conflicts per 4096^3 dispatch (PMC SQ_LDS_BANK_CONFLICT).
-// Involution; applied identically on the glds source and the ds_read
-// address (both-sides-or-neither, rule 21).
-__device__ inline int swz_k(int row, int k) {
-  return k ^ ((row & 7) << 3);
-}
+// LDS XOR swizzle (guide T2, full byte_off ^= (row&15)<<4 form): a
+// ds_read_b128 16-lane group whose rows are distinct mod 16 becomes
+// conflict-free. On a 128-B row the 4-bit xor's top bit crosses the row
+// boundary, so at chunk granularity the involution decomposes into
+//   lds_row  = row ^ ((row>>3)&1)        (row-parity flip from xor bit 3)
+//   lds_k    = k ^ ((row&7)<<3)          (chunk xor from xor bits 0-2)
+// Evolution measured on this kernel (PMC SQ_LDS_BANK_CONFLICT per 4096^3
+// dispatch): st_16x32 single-bit 12.6M conflicts -> (row&7) 8-slot form
+// ~1073 TF -> this form. Same involution on the glds SOURCE (stage_half
+// inverts it) and the ds_read address (rule 21: both-sides-or-neither).
+// row-parity variant (full (row&15)<<4 xor) measured SLOWER: 881-951 TF vs
+// 1020-1073 for the chunk-only form — the scattered glds source row costs
+// more in fetch than the residual 2-way conflicts. Keep chunk-only.
+__device__ inline int swz_row(int row) { return row; }
+__device__ inline int swz_k(int row, int k) { return k ^ ((row & 7) << 3); }
 
 // Stage one [128][64] half-tile: 1024 16-B slots, 512 threads x 2 passes.
 // LDS destination is lane-linear (slot order == lds address order); the
-// swizzle permutes which global 16-B chunk lands in each slot.
+// swizzle permutes which global 16-B chunk lands in each slot: linear slot
+// (row_l, chunk_l) holds global (src_row, chunk_l ^ (src_row&7)) where
+// src_row = row_l ^ ((row_l>>3)&1) (self-inverse row map).
 __device__ inline void stage_half(const __hip_bfloat16* gbase, int ldk,
                                   __hip_bfloat16* half_base) {
   const int t = threadIdx.x;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     int slot = pass * THREADS + t;
-    int row = slot >> 3;
+    int row_l = slot >> 3;
     int chunk = (slot & 7) * 8;
-    glds16(gbase + row * ldk + swz_k(row, chunk), half_base + row * BK + chunk);
+    int src_row = swz_row(row_l);
+    glds16(gbase + src_row * ldk + swz_k(src_row, chunk),
+           half_base + row_l * BK + chunk);
   }
 }
 
 __device__ inline bf16x8 read_frag(const __hip_bfloat16* half_base, int row,
                                    int kbase) {
-  return *reinterpret_cast<const bf16x8*>(half_base + row * BK +
+  return *reinterpret_cast<const bf16x8*>(half_base + swz_row(row) * BK +
                                           swz_k(row, kbase));
 }
 
