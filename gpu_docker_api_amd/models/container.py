@@ -28,6 +28,9 @@ class ContainerRun(BaseModel):
     gpu_count: int = Field(0, alias="gpuCount")
     cpu_count: int = Field(0, alias="cpuCount")
     memory: str = ""
+    # extension: require this much FREE HBM on every allocated GPU
+    # (e.g. "200GB") — the reference can only count whole GPUs
+    gpu_memory: str = Field("", alias="gpuMemory")
     binds: List[Bind] = Field(default_factory=list)
     env: List[str] = Field(default_factory=list)
     cmd: List[str] = Field(default_factory=list)

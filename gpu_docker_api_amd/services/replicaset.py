@@ -135,7 +135,8 @@ class ReplicaSetService:
 
         async with Saga("run") as saga:
             if req.gpu_count > 0:
-                uuids = self.gpu.apply(req.gpu_count)
+                min_hbm = to_bytes(req.gpu_memory) if req.gpu_memory else 0
+                uuids = self.gpu.apply(req.gpu_count, min_free_hbm=min_hbm)
                 saga.push("gpu", lambda: self.gpu.restore(uuids))
                 spec.gpu_uuids = uuids
             else:

@@ -88,6 +88,36 @@ def test_gpu_min_free_hbm_filter(run):
     run(main())
 
 
+def test_run_with_gpu_memory_floor(tmp_path, run):
+    import sys
+
+    sys.path.insert(0, __file__.rsplit("/", 1)[0])
+    from helpers import make_daemon
+
+    from gpu_docker_api_amd.models import ContainerRun
+
+    async def main():
+        d = await make_daemon(tmp_path)
+        # mark GPU 0..6 nearly full; only GPU 7 has 100GB free
+        inv = d.gpu.inventory
+        for i in range(7):
+            inv._used[i] = 250 * 1024**3
+        d.gpu.gpus = inv.enumerate()
+        out = await d.replicaset.run_gpu_container(
+            ContainerRun(
+                image_name="img",
+                replica_set_name="picky",
+                gpu_count=1,
+                gpu_memory="100GB",
+            )
+        )
+        st = await d.runtime.inspect(out["name"])
+        assert st.gpu_uuids == ["MockMI355X-7"]
+        await d.stop()
+
+    run(main())
+
+
 def test_cpu_apply_lowest_free_sorted(run):
     async def main():
         store = MemoryStore()
