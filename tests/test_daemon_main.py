@@ -1,0 +1,142 @@
+"""End-to-end daemon process tests: the real `python -m gpu_docker_api_amd`
+entry point, and the `--state etcd:<endpoint>` branch against a live (fake)
+etcd gateway served over TCP."""
+import json
+import os
+import socket
+import subprocess
+import sys
+import threading
+import time
+
+import httpx
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _wait_ping(port, timeout=60):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        try:
+            if httpx.get(f"http://127.0.0.1:{port}/ping", timeout=1).status_code == 200:
+                return True
+        except Exception:
+            time.sleep(0.2)
+    return False
+
+
+def test_daemon_main_process(tmp_path):
+    port = _free_port()
+    proc = subprocess.Popen(
+        [
+            sys.executable,
+            "-m",
+            "gpu_docker_api_amd",
+            "--addr",
+            f"127.0.0.1:{port}",
+            "--runtime",
+            "mock",
+            "--inventory",
+            "mock",
+            "--dataDir",
+            str(tmp_path / "data"),
+            "--portRange",
+            "49100-49200",
+            "--logLevel",
+            "warning",
+        ],
+        cwd=ROOT,
+        stdout=subprocess.PIPE,
+        stderr=subprocess.STDOUT,
+        start_new_session=True,
+    )
+    try:
+        assert _wait_ping(port), "daemon did not come up"
+        r = httpx.post(
+            f"http://127.0.0.1:{port}/api/v1/replicaSet",
+            json={"imageName": "img", "replicaSetName": "proc", "gpuCount": 2},
+            timeout=30,
+        ).json()
+        assert r["code"] == 200 and r["data"]["name"] == "proc-1"
+        gpus = httpx.get(f"http://127.0.0.1:{port}/api/v1/resources/gpus", timeout=10).json()
+        assert sum(gpus["data"].values()) == 2
+    finally:
+        proc.terminate()  # exact pid
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+            proc.wait(timeout=5)
+
+
+def test_daemon_against_live_etcd_gateway(tmp_path, run):
+    import uvicorn
+
+    from gpu_docker_api_amd.config import Config
+    from gpu_docker_api_amd.routers.app import Daemon
+    from gpu_docker_api_amd.state.etcd_fake import build_fake_etcd
+
+    etcd_port = _free_port()
+    etcd_app = build_fake_etcd()
+    server = uvicorn.Server(
+        uvicorn.Config(etcd_app, host="127.0.0.1", port=etcd_port, log_level="error")
+    )
+    t = threading.Thread(target=server.run, daemon=True)
+    t.start()
+    deadline = time.time() + 30
+    while time.time() < deadline and not server.started:
+        time.sleep(0.05)
+
+    async def main():
+        from gpu_docker_api_amd.models import ContainerRun
+
+        cfg = Config(
+            state=f"etcd:http://127.0.0.1:{etcd_port}",
+            data_dir=str(tmp_path / "d"),
+            runtime="mock",
+            inventory="mock",
+            copy_engine="python",
+            port_range="49300-49400",
+        )
+        d = Daemon(cfg)
+        await d.start()
+        await d.replicaset.run_gpu_container(
+            ContainerRun(image_name="img", replica_set_name="etc", gpu_count=1)
+        )
+        info = await d.replicaset.get_container_info("etc")
+        assert info["containerName"] == "etc-1"
+        await d.queue.drain()
+        await d.stop()
+        # the spec and scheduler state really live in the remote etcd
+        kvs = etcd_app.state.mvcc.range_prefix("/gpu-docker-api/apis/v1/")
+        keys = [kv.key for kv in kvs]
+        assert "/gpu-docker-api/apis/v1/containers/etc" in keys
+        assert "/gpu-docker-api/apis/v1/gpus/gpuStatusMapKey" in keys
+
+        # a second daemon boots from that etcd and sees the allocation
+        d2 = Daemon(
+            Config(
+                state=f"etcd:http://127.0.0.1:{etcd_port}",
+                data_dir=str(tmp_path / "d2"),
+                runtime="mock",
+                inventory="mock",
+                port_range="49300-49400",
+            )
+        )
+        await d2.start()
+        assert sum(d2.gpu.get_gpu_status().values()) == 1
+        assert d2.container_versions.get("etc") == 1
+        await d2.stop()
+
+    run(main())
+    server.should_exit = True
+    t.join(timeout=10)
