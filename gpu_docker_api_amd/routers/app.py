@@ -203,6 +203,39 @@ def _mount(app: FastAPI, d: Daemon) -> None:
     from .response import error, success
     from .codes import Code
 
+    @app.get("/api/v1/events")
+    async def events():
+        """MI355X extension: live state-change stream (SSE). The reference's
+        clients must poll; this streams every container/volume/scheduler
+        state mutation as it commits (memory backend only)."""
+        store = d.store
+        if not hasattr(store, "subscribe"):
+            from .codes import Code as _Code
+            from .response import error as _error
+
+            return _error(_Code.SERVER_BUSY, detail="state backend does not stream events")
+
+        import asyncio
+        import json as _json
+
+        from fastapi.responses import StreamingResponse
+
+        q, unsubscribe = store.subscribe()
+
+        async def gen():
+            try:
+                yield ": connected\n\n"
+                while True:
+                    try:
+                        ev = await asyncio.wait_for(q.get(), timeout=15.0)
+                        yield f"data: {_json.dumps(ev)}\n\n"
+                    except asyncio.TimeoutError:
+                        yield ": keepalive\n\n"
+            finally:
+                unsubscribe()
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
     @app.post("/api/v1/admin/compact")
     async def compact(request: dict | None = None):
         """MI355X extension: discard state history below a revision and

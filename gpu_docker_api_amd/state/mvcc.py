@@ -68,12 +68,24 @@ class MemoryMVCC:
         self._rev = 1            # etcd: store starts at revision 1
         self._compacted = 0      # highest compacted revision (exclusive floor)
         self._hist: Dict[str, List[_Event]] = {}
-        # on_event(key, rev, value, version, create_revision) — value None
+        # observers(key, rev, value, version, create_revision) — value None
         # for tombstones. Version/create_revision ride along so a WAL
         # rewritten after compaction replays with exact per-key numbering.
-        self.on_event: Optional[
-            Callable[[str, int, Optional[str], int, int], None]
-        ] = None
+        # Multiple observers: the WAL plus any event-stream subscribers.
+        self.observers: List[Callable[[str, int, Optional[str], int, int], None]] = []
+
+    @property
+    def on_event(self):  # backward-compat single-observer view
+        return self.observers[0] if self.observers else None
+
+    @on_event.setter
+    def on_event(self, cb) -> None:
+        if cb is not None:
+            self.observers.append(cb)
+
+    def _notify(self, key, rev, value, version, create_rev) -> None:
+        for cb in self.observers:
+            cb(key, rev, value, version, create_rev)
 
     # ------------------------------------------------------------------ info
     @property
@@ -98,8 +110,7 @@ class MemoryMVCC:
                 create_rev, version = last.create_revision, last.version + 1
             events.append(_Event(rev, value, version, create_rev))
             self._rev = rev
-            if self.on_event:
-                self.on_event(key, rev, value, version, create_rev)
+            self._notify(key, rev, value, version, create_rev)
             return KeyValue(key, value, create_rev, rev, version)
 
     def delete(self, key: str) -> int:
@@ -115,8 +126,7 @@ class MemoryMVCC:
             rev = self._rev + 1
             events.append(_Event(rev, None, 0, 0))
             self._rev = rev
-            if self.on_event:
-                self.on_event(key, rev, None, 0, 0)
+            self._notify(key, rev, None, 0, 0)
             return 1
 
     def delete_prefix(self, prefix: str) -> int:
@@ -133,8 +143,7 @@ class MemoryMVCC:
             rev = self._rev + 1
             for k in sorted(live):
                 self._hist[k].append(_Event(rev, None, 0, 0))
-                if self.on_event:
-                    self.on_event(k, rev, None, 0, 0)
+                self._notify(k, rev, None, 0, 0)
             self._rev = rev
             return len(live)
 

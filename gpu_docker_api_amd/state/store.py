@@ -64,6 +64,44 @@ class MemoryStore(StateStore):
         if wal_path:
             self._wal = Wal(wal_path, fsync=fsync)
             self._wal.attach(self.mvcc)
+        # live event-stream subscribers: (asyncio queue, its loop)
+        self._subscribers: list = []
+        self.mvcc.observers.append(self._fanout)
+
+    def _fanout(self, key: str, rev: int, value, version: int, create_rev: int) -> None:
+        if not self._subscribers:
+            return
+        parts = key.rsplit("/", 2)
+        event = {
+            "type": "delete" if value is None else "put",
+            "key": key,
+            "resource": parts[-2] if len(parts) >= 2 else "",
+            "name": parts[-1],
+            "revision": rev,
+            "version": version,
+        }
+        for q, loop in list(self._subscribers):
+            try:  # observers may fire from any thread; hop to the queue's loop
+                loop.call_soon_threadsafe(q.put_nowait, event)
+            except RuntimeError:
+                pass
+
+    def subscribe(self):
+        """Returns (queue, unsubscribe) for the live state-change stream."""
+        import asyncio
+
+        q: asyncio.Queue = asyncio.Queue(maxsize=1024)
+        loop = asyncio.get_running_loop()
+        entry = (q, loop)
+        self._subscribers.append(entry)
+
+        def unsubscribe() -> None:
+            try:
+                self._subscribers.remove(entry)
+            except ValueError:
+                pass
+
+        return q, unsubscribe
 
     async def put(self, resource: Resource, key: str, value: str) -> None:
         self.mvcc.put(resource_key(resource, key), value)
