@@ -27,6 +27,8 @@ class Config:
     inventory: str = "auto"
     mock_gpus: int = 8
     docker_socket: str = "/var/run/docker.sock"
+    # docker driver: attach GPUs via CDI names instead of direct device nodes
+    use_cdi: bool = False
 
     # container policy (reference hardcodes: services/replicaset.go:67-75)
     rootfs_quota: str = "30G"

@@ -105,6 +105,7 @@ class Daemon:
             rt_kwargs = {
                 "socket_path": cfg.docker_socket,
                 "gpu_resolver": self.gpu.info_by_uuid,
+                "use_cdi": cfg.use_cdi,
             }
         self.runtime = make_runtime(cfg.runtime, **rt_kwargs)
 

@@ -27,10 +27,15 @@ class DockerRuntime(RuntimeDriver):
         socket_path: str = "/var/run/docker.sock",
         gpu_resolver: Optional[GpuResolver] = None,
         api_version: str = "v1.41",
+        use_cdi: bool = False,
     ) -> None:
         self.socket_path = socket_path
         self.api = api_version
         self.gpu_resolver: GpuResolver = gpu_resolver or (lambda _u: None)
+        # CDI mode: emit amd.com/gpu=<index> DeviceRequests for hosts with
+        # CDI specs configured (docker >= 25 / podman); default is direct
+        # /dev/kfd + /dev/dri injection which needs no host-side spec.
+        self.use_cdi = use_cdi
         self._session: Optional[aiohttp.ClientSession] = None
 
     def _sess(self) -> aiohttp.ClientSession:
@@ -71,6 +76,14 @@ class DockerRuntime(RuntimeDriver):
     def _materialize_host_config(self, spec: ContainerSpec) -> Dict[str, Any]:
         hc = dict(spec.host_config)
         uuids = spec.gpu_uuids
+        if self.use_cdi and uuids:
+            ids = []
+            for u in uuids:
+                info = self.gpu_resolver(u)
+                ids.append(f"amd.com/gpu={info.index if info else u}")
+            hc["DeviceRequests"] = [{"Driver": "cdi", "DeviceIDs": ids}]
+            hc.pop("Runtime", None)
+            return hc
         devices = docker_devices_for(uuids, self.gpu_resolver)
         if devices:
             existing = {d.get("PathOnHost") for d in (hc.get("Devices") or [])}

@@ -85,6 +85,31 @@ def _resolver(uuid):
     )
 
 
+def test_create_cdi_mode(tmp_path, run):
+    async def main():
+        fake = FakeDockerd()
+        runner = web.AppRunner(fake.app())
+        await runner.setup()
+        sock = str(tmp_path / "docker.sock")
+        await web.UnixSite(runner, sock).start()
+        rt = DockerRuntime(socket_path=sock, gpu_resolver=_resolver, use_cdi=True)
+        spec = ContainerSpec()
+        spec.config = {"Image": "rocm/dev", "Env": []}
+        spec.host_config = {}
+        spec.container_name = "cdi-1"
+        spec.gpu_uuids = ["MockMI355X-2"]
+        await rt.create(spec)
+        hc = fake.containers["cdi-1"]["HostConfig"]
+        assert hc["DeviceRequests"] == [
+            {"Driver": "cdi", "DeviceIDs": ["amd.com/gpu=2"]}
+        ]
+        assert "Devices" not in hc
+        await rt.close()
+        await runner.cleanup()
+
+    run(main())
+
+
 def test_create_injects_rocm_devices(tmp_path, run):
     async def main():
         fake = FakeDockerd()
