@@ -112,11 +112,11 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
         if not req.cmd:
             return error(Code.INVALID_PARAMS)
         try:
-            stdout = await svc.execute_container(name, req)
+            stdout, rc = await svc.execute_container(name, req)
         except Exception as exc:  # noqa: BLE001
             log_error("replicaSet.execute", exc)
             return error(map_error(exc, Code.CONTAINER_EXECUTE_FAILED))
-        return success({"stdout": stdout})
+        return success({"stdout": stdout, "exitCode": rc})
 
     @r.patch("/{name}")
     async def patch(name: str, request: Request):

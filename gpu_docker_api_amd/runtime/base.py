@@ -93,6 +93,11 @@ class RuntimeDriver:
 
     async def execute(self, name: str, cmd: List[str], workdir: str = "") -> str:
         """Run a command inside the container, return combined output."""
+        out, _rc = await self.execute_rc(name, cmd, workdir)
+        return out
+
+    async def execute_rc(self, name: str, cmd: List[str], workdir: str = ""):
+        """Run a command; returns (combined output, exit code)."""
         raise NotImplementedError
 
     async def commit(self, name: str, image: str, tag: str = "") -> str:

@@ -205,7 +205,7 @@ class DockerRuntime(RuntimeDriver):
             return data.decode(errors="replace")  # TTY mode: raw
         return b"".join(out).decode(errors="replace")
 
-    async def execute(self, name: str, cmd: List[str], workdir: str = "") -> str:
+    async def execute_rc(self, name: str, cmd: List[str], workdir: str = ""):
         body: Dict[str, Any] = {
             "AttachStdout": True,
             "AttachStderr": True,
@@ -218,7 +218,13 @@ class DockerRuntime(RuntimeDriver):
         raw = await self._call(
             "POST", f"/exec/{exec_id}/start", body={"Detach": False, "Tty": False}, raw=True
         )
-        return self._demux_stream(raw or b"")
+        rc = 0
+        try:
+            info = await self._call("GET", f"/exec/{exec_id}/json")
+            rc = int(info.get("ExitCode") or 0)
+        except Exception:  # older daemons: treat as success
+            rc = 0
+        return self._demux_stream(raw or b""), rc
 
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         repo, _, t = image.partition(":")

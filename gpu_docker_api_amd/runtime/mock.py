@@ -110,7 +110,7 @@ class MockRuntime(RuntimeDriver):
     async def list(self, all: bool = True) -> List[ContainerState]:
         return [c for c in self.containers.values() if all or c.running]
 
-    async def execute(self, name: str, cmd: List[str], workdir: str = "") -> str:
+    async def execute_rc(self, name: str, cmd: List[str], workdir: str = ""):
         """Run the command on the host chrooted-by-cwd into the rootfs dir
         (close enough for tests; proc/docker drivers do it for real)."""
         self._maybe_fail("execute")
@@ -126,7 +126,7 @@ class MockRuntime(RuntimeDriver):
             stderr=subprocess.STDOUT,
         )
         out, _ = await proc.communicate()
-        return out.decode(errors="replace")
+        return out.decode(errors="replace"), proc.returncode
 
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         self._maybe_fail("commit")

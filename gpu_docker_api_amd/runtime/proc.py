@@ -328,7 +328,7 @@ class ProcRuntime(RuntimeDriver):
             self._refresh(p)
         return [p.state for p in self._procs.values() if all or p.state.running]
 
-    async def execute(self, name: str, cmd: List[str], workdir: str = "") -> str:
+    async def execute_rc(self, name: str, cmd: List[str], workdir: str = ""):
         p = self._get(name)
         self._refresh(p)
         if not p.state.running:
@@ -347,7 +347,7 @@ class ProcRuntime(RuntimeDriver):
             stderr=subprocess.STDOUT,
         )
         out, _ = await proc.communicate()
-        return out.decode(errors="replace")
+        return out.decode(errors="replace"), proc.returncode
 
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         p = self._get(name)

@@ -535,11 +535,12 @@ class ReplicaSetService:
         except ContainerNotExist:
             pass
 
-    async def execute_container(self, name: str, req: ContainerExecute) -> str:
+    async def execute_container(self, name: str, req: ContainerExecute):
         """POST /{name}/execute (reference ExecuteContainer,
-        replicaset.go:225-265)."""
+        replicaset.go:225-265). Returns (stdout, exit code) — the exit code
+        is an extension (the reference returns stdout only)."""
         vname = self._current_vname(name)
-        return await self.runtime.execute(vname, list(req.cmd), req.work_dir)
+        return await self.runtime.execute_rc(vname, list(req.cmd), req.work_dir)
 
     async def commit_container(self, name: str, req: ContainerCommit) -> str:
         """POST /{name}/commit (reference CommitContainer,

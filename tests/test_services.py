@@ -348,10 +348,14 @@ def test_execute_and_commit(tmp_path, run):
     async def main():
         d = await make_daemon(tmp_path)
         await d.replicaset.run_gpu_container(_run_req(gpus=0, cpus=0))
-        out = await d.replicaset.execute_container(
+        out, rc = await d.replicaset.execute_container(
             "demo", ContainerExecute(cmd=["sh", "-c", "echo hello-$PWD"])
         )
-        assert "hello-" in out
+        assert "hello-" in out and rc == 0
+        _out, rc = await d.replicaset.execute_container(
+            "demo", ContainerExecute(cmd=["sh", "-c", "exit 3"])
+        )
+        assert rc == 3
         image = await d.replicaset.commit_container(
             "demo", ContainerCommit(new_image_name="demo-img")
         )
