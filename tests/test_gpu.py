@@ -218,6 +218,39 @@ def test_proc_runtime_gpu_visibility(tmp_path, run):
     run(main())
 
 
+def test_proc_runtime_cgroup_limits(tmp_path, run):
+    """On the GPU box (root, cgroup v2) the proc runtime must actually
+    apply memory.max / cpuset.cpus."""
+    require_gpu()
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    if not os.path.exists("/sys/fs/cgroup/cgroup.controllers"):
+        pytest.skip("no cgroup v2")
+    from gpu_docker_api_amd.models.etcd import ContainerSpec
+    from gpu_docker_api_amd.runtime.proc import ProcRuntime
+
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=True)
+        spec = ContainerSpec()
+        spec.config = {"Cmd": ["sleep", "30"]}
+        spec.host_config = {}
+        spec.container_name = "cg-1"
+        spec.memory_bytes = 512 * 1024 * 1024
+        await rt.create(spec)
+        await rt.start("cg-1")
+        p = rt._procs["cg-1"]
+        if p.cgroup is None:
+            pytest.skip("cgroup writes not permitted in this container")
+        with open(os.path.join(p.cgroup, "memory.max")) as f:
+            assert f.read().strip() == str(512 * 1024 * 1024)
+        with open(os.path.join(p.cgroup, "cgroup.procs")) as f:
+            assert str(p.state.pid) in f.read().split()
+        await rt.close()
+
+    run(main())
+
+
 def test_bench_short_run():
     """bench.py must emit its JSON line on one GPU within minutes."""
     require_gpu()
