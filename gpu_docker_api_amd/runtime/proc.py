@@ -190,11 +190,34 @@ class ProcRuntime(RuntimeDriver):
             env[k] = v
         return env
 
+    @staticmethod
+    def _enable_controllers(path: str) -> None:
+        """cgroup v2: child limits only work when the parent delegates the
+        controllers (+memory +cpuset in cgroup.subtree_control)."""
+        ctrl_file = os.path.join(path, "cgroup.subtree_control")
+        avail_file = os.path.join(path, "cgroup.controllers")
+        try:
+            with open(avail_file) as f:
+                avail = set(f.read().split())
+            want = [c for c in ("memory", "cpuset") if c in avail]
+            for c in want:  # one at a time: a single EINVAL must not kill both
+                try:
+                    with open(ctrl_file, "w") as f:
+                        f.write(f"+{c}")
+                except OSError:
+                    pass
+        except OSError:
+            pass
+
     def _setup_cgroup(self, name: str, p: _Proc, pid: int) -> None:
         if not self.use_cgroups:
             return
-        cg = os.path.join(CGROUP_ROOT, "gda", name)
+        parent = os.path.join(CGROUP_ROOT, "gda")
+        cg = os.path.join(parent, name)
         try:
+            self._enable_controllers(CGROUP_ROOT)
+            os.makedirs(parent, exist_ok=True)
+            self._enable_controllers(parent)
             os.makedirs(cg, exist_ok=True)
             if p.state.memory > 0:
                 with open(os.path.join(cg, "memory.max"), "w") as f:
