@@ -251,6 +251,32 @@ def test_proc_runtime_cgroup_limits(tmp_path, run):
     run(main())
 
 
+def test_proc_volume_loop_quota_enforced(tmp_path, run):
+    """On the GPU box (root), sized volumes are loop-mounted ext4: writing
+    past the size must fail with ENOSPC — real enforcement, not advisory."""
+    require_gpu()
+    from gpu_docker_api_amd.runtime.proc import ProcRuntime
+
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        vs = await rt.volume_create("q-1", {"size": "64MB"})
+        if vs.options.get("enforced") != "loop":
+            pytest.skip("loop mounts not permitted here")
+        try:
+            with pytest.raises(OSError):
+                with open(os.path.join(vs.mountpoint, "big.bin"), "wb") as f:
+                    f.write(b"x" * (128 * 1024 * 1024))  # 2x the quota
+                    f.flush()
+                    os.fsync(f.fileno())
+            # within quota still works
+            with open(os.path.join(vs.mountpoint, "ok.bin"), "wb") as f:
+                f.write(b"y" * (4 * 1024 * 1024))
+        finally:
+            await rt.volume_remove("q-1")
+
+    run(main())
+
+
 def test_bench_short_run():
     """bench.py must emit its JSON line on one GPU within minutes."""
     require_gpu()
