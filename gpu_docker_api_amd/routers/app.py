@@ -13,7 +13,7 @@ import os
 from contextlib import asynccontextmanager
 from typing import Optional
 
-from fastapi import FastAPI
+from fastapi import FastAPI, Request
 from fastapi.responses import PlainTextResponse
 
 from ..config import Config
@@ -238,14 +238,23 @@ def _mount(app: FastAPI, d: Daemon) -> None:
         return StreamingResponse(gen(), media_type="text/event-stream")
 
     @app.post("/api/v1/admin/compact")
-    async def compact(request: dict | None = None):
+    async def compact(request: Request):
         """MI355X extension: discard state history below a revision and
         shrink the WAL (memory backend only). Destroys rollback targets
         below the point — an explicit operator action."""
+        try:
+            body = await request.json()
+        except Exception:
+            body = {}
+        if not isinstance(body, dict):
+            return error(Code.INVALID_PARAMS)
         store = d.store
         if not hasattr(store, "compact"):
             return error(Code.SERVER_BUSY, detail="state backend does not support compaction")
-        rev = int((request or {}).get("revision", 0))
+        try:
+            rev = int(body.get("revision", 0) or 0)
+        except (TypeError, ValueError):
+            return error(Code.INVALID_PARAMS)
         try:
             result = await store.compact(rev)
         except Exception as exc:  # noqa: BLE001

@@ -3,7 +3,7 @@ with what an MI355X operator actually needs: per-GPU HBM free/used bytes and
 the xGMI adjacency matrix (SURVEY.md §2.4 last row)."""
 from __future__ import annotations
 
-from fastapi import APIRouter
+from fastapi import APIRouter, Request
 
 from ..parallel import CpuScheduler, GpuScheduler, PortScheduler
 from .response import success
@@ -23,11 +23,19 @@ def make_router(gpu: GpuScheduler, cpu: CpuScheduler, ports: PortScheduler) -> A
         return success(gpu.get_detail())
 
     @r.post("/gpus/validate")
-    async def gpus_validate(body: dict | None = None):
+    async def gpus_validate(request: Request):
         """MI355X extension: burn-in the node's FREE GPUs (HBM bandwidth +
         dense bf16 MFMA GEMM) before trusting them with placements. Busy
         GPUs are skipped unless explicitly listed."""
-        body = body or {}
+        try:
+            body = await request.json()
+        except Exception:
+            body = {}
+        if not isinstance(body, dict):
+            from .codes import Code
+            from .response import error
+
+            return error(Code.INVALID_PARAMS)
         requested = body.get("gpus")
         status = gpu.get_gpu_status()
         if requested:
