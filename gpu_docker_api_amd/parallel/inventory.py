@@ -154,6 +154,12 @@ class AmdSmiInventory(GpuInventory):
         return out
 
     def _enumerate_cli(self) -> List[GpuInfo]:
+        """amd-smi CLI fallback. Wire formats (captured from a real MI355X,
+        tests/fixtures/amdsmi_*.json):
+          list:   [{"gpu": 0, "bdf": "0000:23:00.0", "uuid": "...", ...}]
+          static: {"gpu_data": [{"gpu": 0, "asic": {"market_name": ...},
+                   "vram": {"size": {"value": 294896, "unit": "MB"}}, ...}]}
+        """
         try:
             raw = subprocess.run(
                 ["amd-smi", "list", "--json"],
@@ -165,16 +171,50 @@ class AmdSmiInventory(GpuInventory):
             data = json.loads(raw)
         except Exception as exc:
             raise RuntimeError(f"GPU enumeration failed (amdsmi lib and CLI): {exc}") from exc
+        static_by_gpu = {}
+        try:
+            raw = subprocess.run(
+                ["amd-smi", "static", "--json"],
+                capture_output=True,
+                text=True,
+                timeout=60,
+                check=True,
+            ).stdout
+            for g in json.loads(raw).get("gpu_data", []) or []:
+                static_by_gpu[int(g.get("gpu", -1))] = g
+        except Exception:
+            pass  # enumeration works without static enrichment
+        return self.parse_cli_output(data, static_by_gpu)
+
+    @staticmethod
+    def parse_cli_output(
+        data, static_by_gpu: Optional[dict] = None
+    ) -> List[GpuInfo]:
+        static_by_gpu = static_by_gpu or {}
         out: List[GpuInfo] = []
         items = data if isinstance(data, list) else data.get("gpu", []) or []
         for i, item in enumerate(items):
+            idx = int(item.get("gpu", i))
             bdf = str(item.get("bdf", ""))
             render, card = _bdf_to_drm_nodes(bdf) if bdf else ("", "")
+            name = "AMD Instinct MI355X"
+            total = MI355X_HBM_BYTES
+            st = static_by_gpu.get(idx) or {}
+            asic = st.get("asic") or {}
+            if asic.get("market_name"):
+                name = str(asic["market_name"])
+            vram = (st.get("vram") or {}).get("size") or {}
+            if isinstance(vram, dict) and vram.get("value"):
+                unit = str(vram.get("unit", "MB")).upper()
+                factor = {"KB": 1024, "MB": 1024**2, "GB": 1024**3}.get(unit, 1024**2)
+                total = int(vram["value"]) * factor
             out.append(
                 GpuInfo(
-                    index=int(item.get("gpu", i)),
+                    index=idx,
                     uuid=str(item.get("uuid", f"GPU-{i}")),
                     bdf=bdf,
+                    name=name,
+                    vram_total=total,
                     render_node=render,
                     card_node=card,
                 )
