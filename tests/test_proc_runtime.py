@@ -104,16 +104,28 @@ def test_pause_unpause_signals(tmp_path, run):
         rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
         await rt.create(_spec("c-1", cmd=["sleep", "30"]))
         await rt.start("c-1")
+        import asyncio
+
         st = await rt.inspect("c-1")
+
+        async def wait_state(pid, want, timeout=3.0):
+            # signal delivery is asynchronous: poll /proc until it lands
+            deadline = asyncio.get_event_loop().time() + timeout
+            state = "?"
+            while asyncio.get_event_loop().time() < deadline:
+                with open(f"/proc/{pid}/stat") as f:
+                    state = f.read().split(") ")[1].split()[0]
+                if state in want:
+                    return state
+                await asyncio.sleep(0.02)
+            return state
+
         await rt.pause("c-1")
-        # process group should be SIGSTOPped
-        with open(f"/proc/{st.pid}/stat") as f:
-            state = f.read().split(") ")[1].split()[0]
+        state = await wait_state(st.pid, ("T", "t"))
         assert state in ("T", "t"), f"expected stopped, got {state}"
         assert (await rt.inspect("c-1")).paused
         await rt.unpause("c-1")
-        with open(f"/proc/{st.pid}/stat") as f:
-            state = f.read().split(") ")[1].split()[0]
+        state = await wait_state(st.pid, ("S", "R"))
         assert state in ("S", "R"), f"expected running, got {state}"
         await rt.close()
 
