@@ -28,6 +28,8 @@ def parse_args(argv=None) -> Config:
     p.add_argument("--inventory", default="auto", choices=["auto", "amdsmi", "mock"])
     p.add_argument("--mockGpus", type=int, default=8, dest="mock_gpus")
     p.add_argument("--dockerSocket", default="/var/run/docker.sock", dest="docker_socket")
+    p.add_argument("--loopVolumes", action="store_true", dest="loop_volumes",
+                   help="proc runtime: enforce sized volumes via loop-mounted ext4")
     p.add_argument("--cdi", action="store_true", dest="use_cdi",
                    help="docker driver: attach GPUs via CDI (amd.com/gpu=N) instead of device nodes")
     p.add_argument("--copyEngine", default="auto", choices=["auto", "iouring", "tar", "python"], dest="copy_engine")
@@ -47,6 +49,7 @@ def parse_args(argv=None) -> Config:
         mock_gpus=a.mock_gpus,
         docker_socket=a.docker_socket,
         use_cdi=a.use_cdi,
+        loop_volumes=a.loop_volumes,
         copy_engine=a.copy_engine,
         run_xgmi_probe=a.run_xgmi_probe,
         run_rccl_smoke=a.run_rccl_smoke,

@@ -35,6 +35,10 @@ class Config:
     shm_size_bytes: int = 256 * 1024**3
     restart_policy: str = "unless-stopped"
 
+    # proc runtime: enforce sized volumes via loop-mounted ext4 images
+    # (opt-in: mounts persist beyond the daemon process)
+    loop_volumes: bool = False
+
     # copy engine: auto | iouring | tar | python
     copy_engine: str = "auto"
 

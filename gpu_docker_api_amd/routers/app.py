@@ -98,6 +98,7 @@ class Daemon:
             rt_kwargs = {
                 "base_dir": os.path.join(cfg.data_dir, "procrt"),
                 "gpu_resolver": self.gpu.info_by_uuid,
+                "loop_volumes": cfg.loop_volumes,
             }
         elif cfg.runtime == "mock":
             rt_kwargs = {"base_dir": os.path.join(cfg.data_dir, "mockrt")}

@@ -65,12 +65,16 @@ class ProcRuntime(RuntimeDriver):
         base_dir: str = "",
         gpu_resolver: Optional[GpuResolver] = None,
         use_cgroups: bool = True,
+        loop_volumes: bool = False,
     ) -> None:
         self.base = base_dir or os.path.join(os.getcwd(), ".state", "procrt")
         for sub in ("containers", "volumes", "images"):
             os.makedirs(os.path.join(self.base, sub), exist_ok=True)
         self.gpu_resolver: GpuResolver = gpu_resolver or (lambda _u: None)
         self.use_cgroups = use_cgroups
+        # loop-device quota enforcement is OPT-IN: mounts outlive the
+        # process, so only long-lived daemons should create them
+        self.loop_volumes = loop_volumes
         self._procs: Dict[str, _Proc] = {}
         self.volumes: Dict[str, VolumeState] = {}
         self._supervisor: Optional[asyncio.Task] = None
@@ -411,7 +415,7 @@ class ProcRuntime(RuntimeDriver):
     async def _try_loop_volume(self, vdir: str, mp: str, size_str: str) -> bool:
         from ..models.memory import to_bytes
 
-        if not os.path.exists("/dev/loop-control"):
+        if not self.loop_volumes or not os.path.exists("/dev/loop-control"):
             return False
         try:
             size = to_bytes(size_str)

@@ -258,7 +258,7 @@ def test_proc_volume_loop_quota_enforced(tmp_path, run):
     from gpu_docker_api_amd.runtime.proc import ProcRuntime
 
     async def main():
-        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False, loop_volumes=True)
         vs = await rt.volume_create("q-1", {"size": "64MB"})
         if vs.options.get("enforced") != "loop":
             pytest.skip("loop mounts not permitted here")
