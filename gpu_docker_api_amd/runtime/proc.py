@@ -298,6 +298,13 @@ class ProcRuntime(RuntimeDriver):
             if p.popen.poll() is None:
                 self._signal_group(p, signal.SIGKILL)
                 p.popen.wait(timeout=5)
+        if p.popen is not None:
+            # the leader is gone; sweep any group stragglers (orphaned
+            # children). pgid == leader pid (start_new_session).
+            try:
+                os.killpg(p.popen.pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                pass
         p.state.running, p.state.paused, p.state.status = False, False, "exited"
         p.state.pid = 0
 
