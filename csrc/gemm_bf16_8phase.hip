@@ -190,6 +190,10 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
         bfrag[0][fn][kk] = read_frag(Bs0, brow + fn * 16, kk * 32 + fk);
     }
     if (4 * t + 7 < 4 * T) stage_h(4 * t + 7);
+    // 12 ds_reads issued this phase: partial wait before the barrier lets
+    // the first reads land while the rest fly (guide template's optional
+    // lgkmcnt(8) line)
+    asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     wait_lgkm0_fence();
     __builtin_amdgcn_s_setprio(1);
