@@ -95,6 +95,7 @@ def validate_gpus(indices: Optional[list] = None, size: int = 4096, iters: int =
     ext = load_ext()
     n = ext.device_count()
     idx = list(indices) if indices else list(range(n))
+    size = max(256, (size // 256) * 256)  # kernel tile constraints
     report = {"gpus": []}
     for i in idx:
         if i < 0 or i >= n:

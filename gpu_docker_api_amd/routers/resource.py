@@ -49,10 +49,18 @@ def make_router(gpu: GpuScheduler, cpu: CpuScheduler, ports: PortScheduler) -> A
                 g.index for g in gpu.gpus if status.get(g.uuid, 1) == 0
             ]
         try:
+            size = int(body.get("size", 4096))
+            iters = int(body.get("iters", 5))
+        except (TypeError, ValueError):
+            from .codes import Code
+            from .response import error
+
+            return error(Code.INVALID_PARAMS)
+        try:
             from ..ops import hipcore
 
             report = await hipcore.validate_gpus_async(
-                idx, int(body.get("size", 4096)), int(body.get("iters", 5))
+                idx, max(256, min(size, 16384)), max(1, min(iters, 50))
             )
         except Exception as exc:  # noqa: BLE001 — no GPU / extension absent
             from .codes import Code
