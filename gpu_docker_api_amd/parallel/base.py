@@ -37,4 +37,4 @@ class BaseScheduler:
 
     async def persist(self) -> None:
         """Synchronous persist (shutdown path)."""
-        await self._store.put(self.resource, self.state_key, self.serialize())
+        await self._store.put_ephemeral(self.resource, self.state_key, self.serialize())

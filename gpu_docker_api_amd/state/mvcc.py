@@ -99,7 +99,14 @@ class MemoryMVCC:
             return self._compacted
 
     # ----------------------------------------------------------------- write
-    def put(self, key: str, value: str) -> KeyValue:
+    def put(self, key: str, value: str, retain_history: bool = True) -> KeyValue:
+        """Write a key. ``retain_history=False`` replaces the key's previous
+        put event instead of appending — for high-churn singleton state
+        (scheduler bitmaps, version maps) whose history nobody walks; without
+        it those keys grow the store without bound (etcd has the same issue
+        and answers with compaction). Revision/version counters still advance
+        so readers can't tell the difference at the head.
+        """
         with self._lock:
             rev = self._rev + 1
             events = self._hist.setdefault(key, [])
@@ -108,6 +115,8 @@ class MemoryMVCC:
                 create_rev, version = rev, 1
             else:
                 create_rev, version = last.create_revision, last.version + 1
+                if not retain_history:
+                    events.pop()  # collapse: at most one live event retained
             events.append(_Event(rev, value, version, create_rev))
             self._rev = rev
             self._notify(key, rev, value, version, create_rev)

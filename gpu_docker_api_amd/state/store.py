@@ -28,6 +28,12 @@ class StateStore:
     async def put(self, resource: Resource, key: str, value: str) -> None:
         raise NotImplementedError
 
+    async def put_ephemeral(self, resource: Resource, key: str, value: str) -> None:
+        """Write WITHOUT retaining history — for the high-churn singleton
+        state keys (scheduler bitmaps, version maps). Default: plain put
+        (real etcd keeps revisions until compacted, as with the reference)."""
+        await self.put(resource, key, value)
+
     async def get(self, resource: Resource, key: str) -> KeyValue:
         """Raises NotExistInStore when absent."""
         raise NotImplementedError
@@ -105,6 +111,9 @@ class MemoryStore(StateStore):
 
     async def put(self, resource: Resource, key: str, value: str) -> None:
         self.mvcc.put(resource_key(resource, key), value)
+
+    async def put_ephemeral(self, resource: Resource, key: str, value: str) -> None:
+        self.mvcc.put(resource_key(resource, key), value, retain_history=False)
 
     async def get(self, resource: Resource, key: str) -> KeyValue:
         return self.mvcc.get(resource_key(resource, key))

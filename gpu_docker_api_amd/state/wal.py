@@ -27,6 +27,10 @@ class Wal:
         os.makedirs(os.path.dirname(os.path.abspath(path)) or ".", exist_ok=True)
         self._f = None
         self._store: Optional[MemoryMVCC] = None
+        self._records_since_check = 0
+        # auto-rewrite threshold: ephemeral puts collapse in-memory history,
+        # so a rewrite shrinks the log back to retained events
+        self.max_bytes = 64 * 1024 * 1024
 
     def attach(self, store: MemoryMVCC) -> None:
         """Replay existing log into ``store``, then record its future events."""
@@ -67,6 +71,14 @@ class Wal:
         self._f.flush()
         if self.fsync:
             os.fsync(self._f.fileno())
+        self._records_since_check += 1
+        if self._records_since_check >= 8192:
+            self._records_since_check = 0
+            try:
+                if os.path.getsize(self.path) > self.max_bytes:
+                    self.rewrite()
+            except OSError:
+                pass
 
     def rewrite(self) -> int:
         """Rewrite the log with only the store's retained events. Returns the

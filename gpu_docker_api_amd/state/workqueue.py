@@ -67,7 +67,10 @@ class WorkQueue:
 
     async def _apply(self, item: PutKeyValue | DelKey) -> None:
         if isinstance(item, PutKeyValue):
-            await self.store.put(item.resource, item.key, item.value)
+            # everything routed through the queue is singleton scheduler /
+            # version-map state: ephemeral (no history retention) — a 250
+            # ops/s daemon would otherwise grow the store without bound
+            await self.store.put_ephemeral(item.resource, item.key, item.value)
         else:
             await self.store.delete(item.resource, item.key)
 

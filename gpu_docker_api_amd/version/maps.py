@@ -43,7 +43,7 @@ class _PersistedMap:
             self._queue.put(self.resource, self._key, self.serialize())
 
     async def persist(self) -> None:
-        await self._store.put(self.resource, self._key, self.serialize())
+        await self._store.put_ephemeral(self.resource, self._key, self.serialize())
 
 
 class VersionMap(_PersistedMap):
