@@ -63,10 +63,15 @@ class _Event:
 class MemoryMVCC:
     """The store. Mutations may be observed via ``on_event`` (used by the WAL)."""
 
-    def __init__(self) -> None:
+    def __init__(self, prune_dead_lifetimes: bool = False) -> None:
         self._lock = threading.RLock()
         self._rev = 1            # etcd: store starts at revision 1
         self._compacted = 0      # highest compacted revision (exclusive floor)
+        # when True, re-creating a deleted key drops the dead lifetime's
+        # events (only at-revision reads of dead lifetimes could see them;
+        # nothing in the daemon does). Bounds churn like delete+recreate
+        # loops; etcd's answer to the same growth is compaction.
+        self.prune_dead_lifetimes = prune_dead_lifetimes
         self._hist: Dict[str, List[_Event]] = {}
         # observers(key, rev, value, version, create_revision) — value None
         # for tombstones. Version/create_revision ride along so a WAL
@@ -113,6 +118,8 @@ class MemoryMVCC:
             last = events[-1] if events else None
             if last is None or last.value is None:
                 create_rev, version = rev, 1
+                if last is not None and self.prune_dead_lifetimes:
+                    events[:] = [last]  # keep only the tombstone
             else:
                 create_rev, version = last.create_revision, last.version + 1
                 if not retain_history:

@@ -65,7 +65,7 @@ class StateStore:
 
 class MemoryStore(StateStore):
     def __init__(self, wal_path: Optional[str] = None, fsync: bool = False) -> None:
-        self.mvcc = MemoryMVCC()
+        self.mvcc = MemoryMVCC(prune_dead_lifetimes=True)
         self._wal: Optional[Wal] = None
         if wal_path:
             self._wal = Wal(wal_path, fsync=fsync)

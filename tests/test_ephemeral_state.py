@@ -48,3 +48,22 @@ def test_store_ephemeral_and_wal_rewrite(tmp_path, run):
         await st2.close()
 
     run(main())
+
+
+def test_dead_lifetime_pruning_bounds_churn(run):
+    async def main():
+        s = MemoryStore()
+        # 500 create/patch/delete cycles of the same name (soak pattern)
+        for i in range(500):
+            await s.put(Resource.CONTAINERS, "churn", f"v1-{i}")
+            await s.put(Resource.CONTAINERS, "churn", f"v2-{i}")
+            await s.delete(Resource.CONTAINERS, "churn")
+        key = "/gpu-docker-api/apis/v1/containers/churn"
+        # bounded: tombstone + nothing else pending (dead lifetimes pruned)
+        assert len(s.mvcc._hist[key]) <= 3
+        # live history still full within a lifetime
+        await s.put(Resource.CONTAINERS, "churn", "a")
+        await s.put(Resource.CONTAINERS, "churn", "b")
+        assert [kv.value for kv in await s.history(Resource.CONTAINERS, "churn")] == ["b", "a"]
+
+    run(main())
