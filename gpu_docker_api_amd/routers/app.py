@@ -8,6 +8,7 @@ synchronously (main.go:139-154) — plus closes the runtime and the WAL.
 """
 from __future__ import annotations
 
+import asyncio
 import logging
 import os
 from contextlib import asynccontextmanager
@@ -133,6 +134,27 @@ class Daemon:
         )
         if cfg.run_rccl_smoke:
             await self._run_rccl_smoke()
+        self._janitor = asyncio.get_running_loop().create_task(self._janitor_loop())
+
+    async def _janitor_loop(self) -> None:
+        """Periodic housekeeping: glibc keeps freed churn memory in arenas
+        (observed ~5 KB/cycle RSS creep under sustained load on a 256-core
+        box while the Python heap stays bounded per tracemalloc);
+        malloc_trim returns it to the kernel."""
+        import ctypes
+        import gc
+
+        try:
+            libc = ctypes.CDLL("libc.so.6")
+        except OSError:
+            return
+        while True:
+            await asyncio.sleep(60.0)
+            gc.collect()
+            try:
+                libc.malloc_trim(0)
+            except Exception:
+                return
 
     async def _run_probe(self) -> Optional[dict]:
         """Run the native xGMI bandwidth probe (csrc -> ops.hipcore)."""
@@ -154,6 +176,9 @@ class Daemon:
             log.warning("RCCL smoke unavailable: %s", exc)
 
     async def stop(self) -> None:
+        if getattr(self, "_janitor", None) is not None:
+            self._janitor.cancel()
+            self._janitor = None
         # drain async writes, then persist everything synchronously
         if self.queue is not None:
             await self.queue.close()
