@@ -290,6 +290,9 @@ class ProcRuntime(RuntimeDriver):
         self._refresh(p)
         if p.popen is not None and p.popen.poll() is None:
             self._signal_group(p, signal.SIGTERM)
+            # a paused (SIGSTOPped) group keeps SIGTERM pending forever:
+            # continue it so termination can be delivered (docker semantics)
+            self._signal_group(p, signal.SIGCONT)
             deadline = time.monotonic() + timeout
             delay = 0.0002  # most processes die in <1 ms; back off geometrically
             while time.monotonic() < deadline and p.popen.poll() is None:

@@ -208,3 +208,31 @@ def test_stop_kills_whole_process_group(tmp_path, run):
         await rt.close()
 
     run(main())
+
+
+def test_stop_kills_paused_container(tmp_path, run):
+    """SIGTERM stays pending on a SIGSTOPped group — stop must SIGCONT it
+    (found as hours-old frozen sleeps leaked by earlier test runs)."""
+    import asyncio
+
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        await rt.create(_spec("c-1", cmd=["sleep", "30"]))
+        await rt.start("c-1")
+        await rt.pause("c-1")
+        st = await rt.inspect("c-1")
+        pid = st.pid
+        await rt.stop("c-1", timeout=3)
+        for _ in range(100):
+            try:
+                os.kill(pid, 0)
+            except ProcessLookupError:
+                break
+            await asyncio.sleep(0.02)
+        else:
+            # zombie state also counts as dead for this purpose
+            with open(f"/proc/{pid}/stat") as f:
+                assert f.read().split(") ")[1].split()[0] == "Z"
+        await rt.close()
+
+    run(main())
