@@ -76,21 +76,21 @@ class Ring {
     fd_ = sys_io_uring_setup(kEntries, &params_);
     if (fd_ < 0) return false;
 
-    size_t sring_sz = params_.sq_off.array + params_.sq_entries * sizeof(unsigned);
-    size_t cring_sz =
-        params_.cq_off.cqes + params_.cq_entries * sizeof(io_uring_cqe);
+    sring_sz_ = params_.sq_off.array + params_.sq_entries * sizeof(unsigned);
+    cring_sz_ = params_.cq_off.cqes + params_.cq_entries * sizeof(io_uring_cqe);
     bool single_mmap = params_.features & IORING_FEAT_SINGLE_MMAP;
-    if (single_mmap && cring_sz > sring_sz) sring_sz = cring_sz;
+    if (single_mmap && cring_sz_ > sring_sz_) sring_sz_ = cring_sz_;
 
-    sq_ptr_ = mmap(nullptr, sring_sz, PROT_READ | PROT_WRITE,
+    sq_ptr_ = mmap(nullptr, sring_sz_, PROT_READ | PROT_WRITE,
                    MAP_SHARED | MAP_POPULATE, fd_, IORING_OFF_SQ_RING);
     if (sq_ptr_ == MAP_FAILED) return false;
     cq_ptr_ = single_mmap
                   ? sq_ptr_
-                  : mmap(nullptr, cring_sz, PROT_READ | PROT_WRITE,
+                  : mmap(nullptr, cring_sz_, PROT_READ | PROT_WRITE,
                          MAP_SHARED | MAP_POPULATE, fd_, IORING_OFF_CQ_RING);
     if (cq_ptr_ == MAP_FAILED) return false;
-    sqes_ = (io_uring_sqe*)mmap(nullptr, params_.sq_entries * sizeof(io_uring_sqe),
+    sqes_sz_ = params_.sq_entries * sizeof(io_uring_sqe);
+    sqes_ = (io_uring_sqe*)mmap(nullptr, sqes_sz_,
                                 PROT_READ | PROT_WRITE, MAP_SHARED | MAP_POPULATE,
                                 fd_, IORING_OFF_SQES);
     if (sqes_ == MAP_FAILED) return false;
@@ -147,6 +147,13 @@ class Ring {
   void wait_one() { submit(1); }
 
   ~Ring() {
+    // closing the fd does NOT unmap the rings — without these munmaps a
+    // long-lived daemon leaks ~2 pages per copy_tree call (found as ~8.6
+    // KB/cycle RSS creep in a 15-minute churn soak)
+    if (sqes_ != nullptr && sqes_ != MAP_FAILED) munmap(sqes_, sqes_sz_);
+    if (cq_ptr_ != nullptr && cq_ptr_ != MAP_FAILED && cq_ptr_ != sq_ptr_)
+      munmap(cq_ptr_, cring_sz_);
+    if (sq_ptr_ != nullptr && sq_ptr_ != MAP_FAILED) munmap(sq_ptr_, sring_sz_);
     if (fd_ >= 0) close(fd_);
   }
 
@@ -156,6 +163,9 @@ class Ring {
   io_uring_params params_{};
   void* sq_ptr_ = nullptr;
   void* cq_ptr_ = nullptr;
+  size_t sring_sz_ = 0;
+  size_t cring_sz_ = 0;
+  size_t sqes_sz_ = 0;
   io_uring_sqe* sqes_ = nullptr;
   std::atomic<unsigned>* sq_tail_ = nullptr;
   unsigned sq_mask_ = 0;
