@@ -154,6 +154,12 @@ vol = typer.Typer(help="volumes")
 app.add_typer(vol, name="volume")
 
 
+@vol.command("ls")
+def vol_ls():
+    with _client() as c:
+        _show(c.get("/api/v1/volumes"))
+
+
 @vol.command("create")
 def vol_create(name: str, size: str = typer.Option("", "--size", "-s")):
     with _client() as c:
@@ -200,6 +206,37 @@ def res_cpus():
 def res_ports():
     with _client() as c:
         _show(c.get("/api/v1/resources/ports"))
+
+
+@res.command("validate")
+def res_validate(
+    size: int = typer.Option(4096, "--size"),
+    iters: int = typer.Option(5, "--iters"),
+):
+    """Burn-in free GPUs (HBM bandwidth + bf16 MFMA TFLOPS)."""
+    with _client() as c:
+        _show(
+            c.post(
+                "/api/v1/resources/gpus/validate",
+                json={"size": size, "iters": iters},
+                timeout=600,
+            )
+        )
+
+
+@app.command()
+def events():
+    """Tail the live state-change stream (SSE); Ctrl-C to stop."""
+    headers = {}
+    apikey = os.environ.get("APIKEY", "")
+    if apikey:
+        headers["Authorization"] = f"Bearer {apikey}"
+    with httpx.stream(
+        "GET", f"{DEFAULT_ADDR}/api/v1/events", headers=headers, timeout=None
+    ) as resp:
+        for line in resp.iter_lines():
+            if line.startswith("data: "):
+                typer.echo(line[6:])
 
 
 def main() -> None:

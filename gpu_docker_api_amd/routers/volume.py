@@ -49,6 +49,16 @@ def make_router(svc: VolumeService) -> APIRouter:
             return error(map_error(exc, Code.VOLUME_CREATE_FAILED, volume=True))
         return success(data)
 
+    @r.get("")
+    async def list_all():
+        """MI355X extension: list all volumes."""
+        try:
+            data = await svc.list_volumes()
+        except Exception as exc:  # noqa: BLE001
+            log_error("volume.list", exc)
+            return error(map_error(exc, Code.VOLUME_GET_INFO_FAILED, volume=True))
+        return success(data)
+
     @r.patch("/{name}/size")
     async def patch_size(name: str, request: Request):
         req = await _parse(request, VolumeSize)

@@ -171,6 +171,27 @@ class VolumeService:
             await self.store.delete(Resource.VOLUMES, name)
 
     # ------------------------------------------------------------------- info
+    async def list_volumes(self) -> List[Dict]:
+        """All volumes with spec + live state (extension)."""
+        out = []
+        for name, version in sorted(self.versions.snapshot().items()):
+            kv = await self.store.get_or_none(Resource.VOLUMES, name)
+            if kv is None:
+                continue
+            spec = VolumeSpec.deserialize(kv.value)
+            vs = await self.runtime.volume_inspect(spec.name)
+            out.append(
+                {
+                    "name": name,
+                    "volumeName": spec.name,
+                    "version": version,
+                    "size": spec.size,
+                    "mountpoint": vs.mountpoint if vs else "",
+                    "present": vs is not None,
+                }
+            )
+        return out
+
     async def get_volume_info(self, name: str) -> Dict:
         return (await self._load_spec(name)).to_dict()
 

@@ -81,6 +81,10 @@ def test_cli_full_lifecycle(live_daemon):
     out = _cli(port, "volume", "create", "vol", "--size", "5GB")
     assert json.loads(out.stdout)["name"] == "vol-1"
 
+    out = _cli(port, "volume", "ls")
+    vols = json.loads(out.stdout)
+    assert vols[0]["name"] == "vol" and vols[0]["size"] == "5GB"
+
     out = _cli(port, "delete", "job")
     assert out.returncode == 0
 
