@@ -158,6 +158,21 @@ def test_create_injects_rocm_devices(tmp_path, run):
         vs = await rt.volume_create("vol-1", {"size": "10GB"})
         assert vs.mountpoint.endswith("vol-1/_data")
 
+        # lifecycle verbs hit the right endpoints with the right params
+        await rt.stop("demo-1", timeout=7)
+        await rt.restart("demo-1", timeout=9)
+        await rt.pause("demo-1")
+        await rt.unpause("demo-1")
+        await rt.remove("demo-1", force=True)
+        await rt.commit("demo-1", "snap:v2")
+        calls = [(m, p, q) for m, p, _b, q in fake.requests]
+        assert ("POST", "/v1.41/containers/demo-1/stop", {"t": "7"}) in calls
+        assert ("POST", "/v1.41/containers/demo-1/restart", {"t": "9"}) in calls
+        assert ("POST", "/v1.41/containers/demo-1/pause", {}) in calls
+        assert ("POST", "/v1.41/containers/demo-1/unpause", {}) in calls
+        assert ("DELETE", "/v1.41/containers/demo-1", {"force": "true", "v": "false"}) in calls
+        assert ("POST", "/v1.41/commit", {"container": "demo-1", "repo": "snap", "tag": "v2"}) in calls
+
         await rt.close()
         await runner.cleanup()
 
