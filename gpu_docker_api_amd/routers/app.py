@@ -81,6 +81,15 @@ class Daemon:
             probe = Topology.load_probe_file(cfg.probe_cache)
         if probe is None and cfg.run_xgmi_probe:
             probe = await self._run_probe()
+            if probe is not None and cfg.probe_cache:
+                # cache the measured link map for the next startup
+                try:
+                    import json as _json
+
+                    with open(cfg.probe_cache, "w") as f:
+                        _json.dump(probe, f)
+                except OSError as exc:
+                    log.warning("could not write probe cache: %s", exc)
         self.gpu = await GpuScheduler.create(self.store, self.queue, inventory, probe=probe)
         self.cpu = await CpuScheduler.create(self.store, self.queue)
         self.ports = await PortScheduler.create(

@@ -215,3 +215,18 @@ def test_merge_map(run):
         assert mm.get("bar-1") is not None
 
     run(main())
+
+
+def test_probe_cache_file_roundtrip(tmp_path):
+    from gpu_docker_api_amd.parallel.topology import Topology
+    import json
+
+    p = tmp_path / "probe.json"
+    p.write_text(json.dumps({"gpus": ["a", "b"], "p2p_gbps": [[0, 140.0], [141.0, 0]]}))
+    probe = Topology.load_probe_file(str(p))
+    t = Topology([[0.0, 50.0], [50.0, 0.0]], ["a", "b"])
+    t.overlay_measured(probe)
+    assert t.bandwidth("a", "b") == 140.0
+    # corrupt file degrades to None, not a crash
+    p.write_text("{not json")
+    assert Topology.load_probe_file(str(p)) is None
