@@ -26,13 +26,14 @@ from ..state.etcd_gateway import EtcdGatewayStore
 from ..state.keys import (
     CONTAINER_MERGE_MAP_KEY,
     CONTAINER_VERSION_MAP_KEY,
+    RELEASED_SET_KEY,
     VOLUME_VERSION_MAP_KEY,
 )
 from ..state.store import MemoryStore, StateStore
 from ..state.workqueue import WorkQueue
 from ..utils.copy import CopyEngine
 from ..utils.timing import METRICS
-from ..version import MergeMap, VersionMap
+from ..version import MergeMap, ReleasedSet, VersionMap
 from . import replicaset as replicaset_router
 from . import resource as resource_router
 from . import volume as volume_router
@@ -99,9 +100,11 @@ class Daemon:
         self.container_versions = VersionMap(self.store, self.queue, CONTAINER_VERSION_MAP_KEY)
         self.volume_versions = VersionMap(self.store, self.queue, VOLUME_VERSION_MAP_KEY)
         self.merges = MergeMap(self.store, self.queue, CONTAINER_MERGE_MAP_KEY)
+        self.released = ReleasedSet(self.store, self.queue, RELEASED_SET_KEY)
         await self.container_versions.load()
         await self.volume_versions.load()
         await self.merges.load()
+        await self.released.load()
 
         rt_kwargs = {}
         if cfg.runtime in ("proc",):
@@ -132,6 +135,7 @@ class Daemon:
             runtime=self.runtime,
             copy_engine=copy_engine,
             cfg=cfg,
+            released=self.released,
         )
         self.volume = VolumeService(
             store=self.store,
@@ -194,7 +198,12 @@ class Daemon:
         for part in (self.gpu, self.cpu, self.ports):
             if part is not None:
                 await part.persist()
-        for part in (self.container_versions, self.volume_versions, self.merges):
+        for part in (
+            self.container_versions,
+            self.volume_versions,
+            self.merges,
+            getattr(self, "released", None),
+        ):
             if part is not None:
                 await part.persist()
         if self.runtime is not None:

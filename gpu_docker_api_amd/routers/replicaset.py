@@ -14,6 +14,7 @@ from ..models import (
 )
 from ..models.memory import parse_size
 from ..services.replicaset import ReplicaSetService
+from ..utils.names import valid_name
 from .codes import Code
 from .errors import log_error, map_error
 from .response import error, success
@@ -25,6 +26,19 @@ def _valid_memory(s: str) -> bool:
         return True
     except Exception:
         return False
+
+
+def _bad_name(name: str):
+    """Reject names that are not strict identifiers before they reach any
+    filesystem join (a name like '..' or 'a/b' in the reference's weaker
+    validation would traverse out of the data dir — see utils/names.py)."""
+    if not name:
+        return error(Code.CONTAINER_NAME_EMPTY)
+    if "-" in name:
+        return error(Code.CONTAINER_NAME_DASH)
+    if not valid_name(name):
+        return error(Code.INVALID_PARAMS, detail=f"invalid replicaSet name: {name!r}")
+    return None
 
 
 def make_router(svc: ReplicaSetService) -> APIRouter:
@@ -43,10 +57,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
             return error(Code.INVALID_PARAMS)
         if not req.image_name:
             return error(Code.IMAGE_NAME_EMPTY)
-        if not req.replica_set_name:
-            return error(Code.CONTAINER_NAME_EMPTY)
-        if "-" in req.replica_set_name:
-            return error(Code.CONTAINER_NAME_DASH)
+        bad = _bad_name(req.replica_set_name)
+        if bad is not None:
+            return bad
         if req.gpu_count < 0:
             return error(Code.GPU_COUNT_GE_ZERO)
         if req.cpu_count < 0:
@@ -73,6 +86,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.get("/{name}")
     async def info(name: str):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         try:
             data = await svc.get_container_info(name)
         except Exception as exc:  # noqa: BLE001
@@ -82,6 +98,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.get("/{name}/history")
     async def history(name: str):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         try:
             data = await svc.get_container_history(name)
         except Exception as exc:  # noqa: BLE001
@@ -91,6 +110,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.post("/{name}/commit")
     async def commit(name: str, request: Request):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         req = await _parse(request, ContainerCommit)
         if req is None:
             return error(Code.INVALID_PARAMS)
@@ -106,6 +128,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.post("/{name}/execute")
     async def execute(name: str, request: Request):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         req = await _parse(request, ContainerExecute)
         if req is None:
             return error(Code.INVALID_PARAMS)
@@ -120,6 +145,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.patch("/{name}")
     async def patch(name: str, request: Request):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         req = await _parse(request, PatchRequest)
         if req is None:
             return error(Code.INVALID_PARAMS)
@@ -142,6 +170,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.patch("/{name}/rollback")
     async def rollback(name: str, request: Request):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         req = await _parse(request, RollbackRequest)
         if req is None:
             return error(Code.INVALID_PARAMS)
@@ -156,6 +187,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.patch("/{name}/stop")
     async def stop(name: str):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         try:
             await svc.stop_container(name)
         except Exception as exc:  # noqa: BLE001
@@ -165,6 +199,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.patch("/{name}/pause")
     async def pause(name: str):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         try:
             await svc.pause_container(name)
         except Exception as exc:  # noqa: BLE001
@@ -174,6 +211,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.patch("/{name}/continue")
     async def cont(name: str):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         try:
             await svc.startup_container(name)
         except Exception as exc:  # noqa: BLE001
@@ -183,6 +223,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.patch("/{name}/restart")
     async def restart(name: str):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         try:
             data = await svc.restart_container(name)
         except Exception as exc:  # noqa: BLE001
@@ -192,6 +235,9 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
 
     @r.delete("/{name}")
     async def delete(name: str):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         try:
             await svc.delete_container(name)
         except Exception as exc:  # noqa: BLE001

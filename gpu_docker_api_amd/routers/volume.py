@@ -7,6 +7,7 @@ from pydantic import ValidationError
 
 from ..models import VolumeCreate, VolumeSize
 from ..models.memory import parse_size
+from ..utils.names import valid_name
 from ..services.volume import VolumeService
 from .codes import Code
 from .errors import log_error, map_error
@@ -29,17 +30,28 @@ def make_router(svc: VolumeService) -> APIRouter:
         except Exception:
             return False
 
+    def _bad_name(name: str):
+        """Strict identifier grammar: the reference's no-dash/no-leading-slash
+        check (volume.go:37-47) still admits '..' and 'a/b', which traverse
+        out of the volumes dir on the proc runtime (utils/names.py)."""
+        if not name:
+            return error(Code.VOLUME_NAME_EMPTY)
+        if "-" in name:
+            return error(Code.VOLUME_NAME_DASH)
+        if name.startswith("/"):
+            return error(Code.VOLUME_NAME_SLASH)
+        if not valid_name(name):
+            return error(Code.INVALID_PARAMS, detail=f"invalid volume name: {name!r}")
+        return None
+
     @r.post("")
     async def create(request: Request):
         req = await _parse(request, VolumeCreate)
         if req is None:
             return error(Code.INVALID_PARAMS)
-        if not req.name:
-            return error(Code.VOLUME_NAME_EMPTY)
-        if "-" in req.name:
-            return error(Code.VOLUME_NAME_DASH)
-        if req.name.startswith("/"):
-            return error(Code.VOLUME_NAME_SLASH)
+        bad = _bad_name(req.name)
+        if bad is not None:
+            return bad
         if req.size and not _valid_size(req.size):
             return error(Code.VOLUME_SIZE_UNIT)
         try:
@@ -61,6 +73,9 @@ def make_router(svc: VolumeService) -> APIRouter:
 
     @r.patch("/{name}/size")
     async def patch_size(name: str, request: Request):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         req = await _parse(request, VolumeSize)
         if req is None:
             return error(Code.INVALID_PARAMS)
@@ -75,6 +90,9 @@ def make_router(svc: VolumeService) -> APIRouter:
 
     @r.delete("/{name}")
     async def delete(name: str, request: Request):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         # ?noall present => keep the store record (reference volume.go:123-129)
         keep = "noall" in request.query_params
         try:
@@ -86,6 +104,9 @@ def make_router(svc: VolumeService) -> APIRouter:
 
     @r.get("/{name}")
     async def info(name: str):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         try:
             data = await svc.get_volume_info(name)
         except Exception as exc:  # noqa: BLE001
@@ -95,6 +116,9 @@ def make_router(svc: VolumeService) -> APIRouter:
 
     @r.get("/{name}/history")
     async def history(name: str):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
         try:
             data = await svc.get_volume_history(name)
         except Exception as exc:  # noqa: BLE001

@@ -33,6 +33,7 @@ import uuid as uuidlib
 from typing import Dict, List, Optional
 
 from ..models.etcd import ContainerSpec
+from ..utils.names import safe_subpath
 from ..xerrors import ContainerExisted, ContainerNotExist, VolumeExisted
 from .base import ContainerState, GpuResolver, RuntimeDriver, VolumeState
 from .devices import visible_device_env
@@ -110,10 +111,16 @@ class ProcRuntime(RuntimeDriver):
 
     # ------------------------------------------------------------------ util
     def _cdir(self, name: str) -> str:
-        return os.path.join(self.base, "containers", name)
+        # containment-checked: these paths feed rmtree
+        return safe_subpath(os.path.join(self.base, "containers"), name)
+
+    def _vdir(self, name: str) -> str:
+        return safe_subpath(os.path.join(self.base, "volumes"), name)
 
     def _image_dir(self, ref: str) -> str:
-        return os.path.join(self.base, "images", ref.replace("/", "_").replace(":", "_"))
+        return safe_subpath(
+            os.path.join(self.base, "images"), ref.replace("/", "_").replace(":", "_")
+        )
 
     def _load_volumes(self) -> None:
         vroot = os.path.join(self.base, "volumes")
@@ -300,7 +307,12 @@ class ProcRuntime(RuntimeDriver):
                 delay = min(delay * 2, 0.02)
             if p.popen.poll() is None:
                 self._signal_group(p, signal.SIGKILL)
-                p.popen.wait(timeout=5)
+                # reap without blocking the event loop: a process stuck in
+                # D-state I/O would otherwise stall every handler for 5 s
+                # 
+                kill_deadline = time.monotonic() + 5
+                while time.monotonic() < kill_deadline and p.popen.poll() is None:
+                    await asyncio.sleep(0.005)
         if p.popen is not None:
             # the leader is gone; sweep any group stragglers (orphaned
             # children). pgid == leader pid (start_new_session).
@@ -460,7 +472,7 @@ class ProcRuntime(RuntimeDriver):
     ) -> VolumeState:
         if name in self.volumes:
             raise VolumeExisted(name)
-        vdir = os.path.join(self.base, "volumes", name)
+        vdir = self._vdir(name)
         mp = os.path.join(vdir, "_data")
         os.makedirs(mp, exist_ok=True)
         opts = dict(driver_opts or {})
@@ -474,7 +486,7 @@ class ProcRuntime(RuntimeDriver):
 
     async def volume_remove(self, name: str, force: bool = True) -> None:
         vs = self.volumes.pop(name, None)
-        vdir = os.path.join(self.base, "volumes", name)
+        vdir = self._vdir(name)
         if vs is not None and vs.options.get("enforced") == "loop":
             proc = await asyncio.create_subprocess_exec(
                 "umount", "-l", vs.mountpoint,

@@ -28,7 +28,7 @@ import asyncio
 import logging
 import os
 import shutil
-from typing import Dict, List, Optional, Set, Tuple
+from typing import Dict, List, Optional, Tuple
 
 from ..config import Config
 from ..models import ContainerCommit, ContainerExecute, ContainerRun, PatchRequest
@@ -40,8 +40,10 @@ from ..state.keys import Resource
 from ..state.store import StateStore
 from ..state.workqueue import WorkQueue
 from ..utils.copy import CopyEngine
+from ..utils.names import safe_subpath
 from ..utils.timing import PhaseTimer
-from ..version import MergeMap, VersionMap
+from ..state.keys import RELEASED_SET_KEY
+from ..version import MergeMap, ReleasedSet, VersionMap
 from ..xerrors import (
     ContainerExisted,
     ContainerNotExist,
@@ -71,6 +73,7 @@ class ReplicaSetService:
         runtime: RuntimeDriver,
         copy_engine: Optional[CopyEngine] = None,
         cfg: Optional[Config] = None,
+        released: Optional[ReleasedSet] = None,
     ) -> None:
         self.store = store
         self.queue = queue
@@ -82,8 +85,13 @@ class ReplicaSetService:
         self.runtime = runtime
         self.copy = copy_engine or CopyEngine()
         self.cfg = cfg or Config()
-        # versioned names whose resources were released by stop_container
-        self._released: Set[str] = set()
+        # versioned names whose resources were released by stop_container —
+        # PERSISTED (write-behind) so a daemon restart cannot forget a stop
+        # and double-book the freed GPUs; Daemon.start
+        # loads it before serving
+        self._released: ReleasedSet = released or ReleasedSet(
+            store, queue, RELEASED_SET_KEY
+        )
 
     # ------------------------------------------------------------ persistence
     async def _persist_spec(self, name: str, spec: ContainerSpec) -> None:
@@ -105,6 +113,19 @@ class ReplicaSetService:
         if v is None:
             raise ContainerNotExist(name)
         return versioned(name, v)
+
+    async def _spec_at_version(self, name: str, version: int) -> ContainerSpec:
+        """The stored spec of history version N, matched on the spec's OWN
+        version field rather than the store's per-key version counter.
+        The counters usually align (one put per container version), but a
+        compensated saga re-put or an etcd backend shared with other
+        writers advances the per-key counter without a container-version
+        bump — matching on content is robust to that (ADVICE r1 #2)."""
+        for kv in await self.store.history(Resource.CONTAINERS, name):
+            spec = ContainerSpec.deserialize(kv.value)
+            if spec.version == version:
+                return spec
+        raise ContainerNotExist(f"{name} version={version}")
 
     # ----------------------------------------------------------------- create
     async def run_gpu_container(self, req: ContainerRun) -> Dict:
@@ -206,7 +227,21 @@ class ReplicaSetService:
             await self.runtime.start(vname)
             if timer:
                 timer.mark("start")
+        # capture the previous spec so a later saga failure (copy/start of
+        # the replacement, old-container removal) restores the CONTAINERS
+        # key too — without this the store would keep the new spec while the
+        # version map reverted, so _load_spec and _current_vname disagree
+        # afterwards
+        prev = await self.store.get_or_none(Resource.CONTAINERS, name)
+
+        async def _undo_persist(prev=prev):
+            if prev is None:
+                await self.store.delete(Resource.CONTAINERS, name)
+            else:
+                await self.store.put(Resource.CONTAINERS, name, prev.value)
+
         await self._persist_spec(name, spec)
+        saga.push("spec", _undo_persist)
         if timer:
             timer.mark("persist")
         return cid, vname
@@ -329,7 +364,9 @@ class ReplicaSetService:
         # — the old container is deleted next anyway; docker's UpperDir
         # belongs to overlayfs, so it is copied.
         if old_state is not None and old_state.upper_dir and os.path.isdir(old_state.upper_dir):
-            merge_path = os.path.join(self.cfg.merges_dir, name, vname_old)
+            # safe_subpath: names are router-validated, but never trust a
+            # join that feeds an rmtree (defense in depth — ADVICE r1 #1)
+            merge_path = safe_subpath(self.cfg.merges_dir, name, vname_old)
             try:
                 if getattr(self.runtime, "owns_rootfs", False):
                     os.makedirs(os.path.dirname(merge_path), exist_ok=True)
@@ -404,8 +441,7 @@ class ReplicaSetService:
             raise ContainerNotExist(name)
         if cur_version == target_version:
             raise NoRollbackRequired(name)
-        kv = await self.store.get_version(Resource.CONTAINERS, name, target_version)
-        target = ContainerSpec.deserialize(kv.value)
+        target = await self._spec_at_version(name, target_version)
         live = await self._load_spec(name)
         vname_old = versioned(name, cur_version)
 
@@ -523,7 +559,7 @@ class ReplicaSetService:
                 self.ports.restore(self._host_ports(st))
         self._released.discard(vname)
         # wipe preserved layers + merge map entries + version map + store key
-        merges_path = os.path.join(self.cfg.merges_dir, name)
+        merges_path = safe_subpath(self.cfg.merges_dir, name)
         await asyncio.get_running_loop().run_in_executor(
             None, lambda: shutil.rmtree(merges_path, ignore_errors=True)
         )

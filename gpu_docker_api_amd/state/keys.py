@@ -48,3 +48,4 @@ USED_PORT_SET_KEY = "usedPortSetKey"
 CONTAINER_VERSION_MAP_KEY = "containerVersionMapKey"
 VOLUME_VERSION_MAP_KEY = "volumeVersionMapKey"
 CONTAINER_MERGE_MAP_KEY = "containerMergeMapKey"
+RELEASED_SET_KEY = "releasedSetKey"

@@ -1,0 +1,51 @@
+"""Resource-name validation and path-containment guards.
+
+The reference validates names only for emptiness and dashes
+(/root/reference/internal/routers/replicaset.go:144-148, volume.go:37-47),
+which lets a name like ``..`` or ``a/b`` reach filesystem joins
+(``merges/<name>``, the proc runtime's container/volume dirs) and escape the
+data directory — ``DELETE /replicaSet/..`` would rmtree the parent of
+merges_dir. Here every name must match a strict identifier grammar at the
+API boundary, and every filesystem join inside the services/runtime is
+additionally realpath-checked against its base directory (defense in depth).
+"""
+from __future__ import annotations
+
+import os
+import re
+
+# First char alnum/underscore (so "." and ".." can never match); dash is
+# excluded everywhere because "-<version>" is the version-suffix separator.
+NAME_RE = re.compile(r"^[A-Za-z0-9_][A-Za-z0-9_.]{0,63}$")
+
+# Versioned names ("<name>-<version>") as produced by the services.
+VERSIONED_NAME_RE = re.compile(r"^[A-Za-z0-9_][A-Za-z0-9_.]{0,63}(-[0-9]+)?$")
+
+
+def valid_name(name: str) -> bool:
+    """True when ``name`` is a safe replicaSet/volume identifier."""
+    return bool(NAME_RE.match(name or ""))
+
+
+def valid_versioned_name(name: str) -> bool:
+    return bool(VERSIONED_NAME_RE.match(name or ""))
+
+
+def safe_subpath(base: str, *parts: str) -> str:
+    """Join ``parts`` under ``base`` and assert the result cannot escape it.
+
+    Raises ValueError when any part is absolute or the resolved path lands
+    outside ``base`` (symlinks in *parts themselves* are not followed — the
+    guard is against traversal in the name components).
+    """
+    for p in parts:
+        if not p or os.path.isabs(p):
+            raise ValueError(f"unsafe path component: {p!r}")
+    joined = os.path.join(base, *parts)
+    base_real = os.path.realpath(base)
+    target_real = os.path.realpath(joined)
+    # strictly below base: equality means the parts normalized away (".")
+    # and the caller would operate on the base dir itself
+    if not target_real.startswith(base_real + os.sep):
+        raise ValueError(f"path escapes {base!r}: {joined!r}")
+    return joined

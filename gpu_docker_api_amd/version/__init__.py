@@ -1,3 +1,3 @@
-from .maps import VersionMap, MergeMap
+from .maps import MergeMap, ReleasedSet, VersionMap
 
-__all__ = ["VersionMap", "MergeMap"]
+__all__ = ["VersionMap", "MergeMap", "ReleasedSet"]
