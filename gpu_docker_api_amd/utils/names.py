@@ -16,19 +16,19 @@ import re
 
 # First char alnum/underscore (so "." and ".." can never match); dash is
 # excluded everywhere because "-<version>" is the version-suffix separator.
-NAME_RE = re.compile(r"^[A-Za-z0-9_][A-Za-z0-9_.]{0,63}$")
+NAME_RE = re.compile(r"[A-Za-z0-9_][A-Za-z0-9_.]{0,63}")
 
 # Versioned names ("<name>-<version>") as produced by the services.
-VERSIONED_NAME_RE = re.compile(r"^[A-Za-z0-9_][A-Za-z0-9_.]{0,63}(-[0-9]+)?$")
+VERSIONED_NAME_RE = re.compile(r"[A-Za-z0-9_][A-Za-z0-9_.]{0,63}(-[0-9]+)?")
 
 
 def valid_name(name: str) -> bool:
     """True when ``name`` is a safe replicaSet/volume identifier."""
-    return bool(NAME_RE.match(name or ""))
+    return bool(NAME_RE.fullmatch(name or ""))  # fullmatch: "$" would accept a trailing newline
 
 
 def valid_versioned_name(name: str) -> bool:
-    return bool(VERSIONED_NAME_RE.match(name or ""))
+    return bool(VERSIONED_NAME_RE.fullmatch(name or ""))
 
 
 def safe_subpath(base: str, *parts: str) -> str:
