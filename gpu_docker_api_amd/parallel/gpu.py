@@ -40,7 +40,12 @@ class GpuScheduler(BaseScheduler):
         super().__init__(store, queue)
         self.inventory = inventory
         self.gpus: List[GpuInfo] = []
-        self.available_gpu_nums = 0
+        # total GPUs on the node — NOT a free count. The reference persists
+        # this as "availableGpuNums" with the same never-decremented meaning
+        # (gpuscheduler.go:32-37); the wire field name is kept for state
+        # compatibility, the Python name says what it is.
+        # The real free-count check happens under the lock in apply().
+        self.node_gpu_count = 0
         self.gpu_status_map: Dict[str, int] = {}
         self.topology: Topology = Topology([], [])
 
@@ -64,10 +69,10 @@ class GpuScheduler(BaseScheduler):
             persisted = data.get("gpuStatusMap") or {}
             # keep persisted bits for GPUs that still exist; new GPUs are free
             self.gpu_status_map = {u: int(persisted.get(u, 0)) for u in uuids}
-            self.available_gpu_nums = len(uuids)
+            self.node_gpu_count = len(uuids)
         else:
             self.gpu_status_map = {u: 0 for u in uuids}
-            self.available_gpu_nums = len(uuids)
+            self.node_gpu_count = len(uuids)
             await self.persist()
         return self
 
@@ -76,7 +81,7 @@ class GpuScheduler(BaseScheduler):
         with self._lock:
             return json.dumps(
                 {
-                    "availableGpuNums": self.available_gpu_nums,
+                    "availableGpuNums": self.node_gpu_count,
                     "gpuStatusMap": self.gpu_status_map,
                 },
                 separators=(",", ":"),
@@ -84,11 +89,11 @@ class GpuScheduler(BaseScheduler):
 
     def apply(self, num: int, min_free_hbm: int = 0) -> List[str]:
         """Allocate ``num`` GPUs, topology-aware. Returns their UUIDs."""
-        if num <= 0 or num > self.available_gpu_nums:
-            raise GpuNotEnough(
-                f"requested {num}, node has {self.available_gpu_nums}"
-            )
         with self._lock:
+            if num <= 0 or num > self.node_gpu_count:
+                raise GpuNotEnough(
+                    f"requested {num}, node has {self.node_gpu_count}"
+                )
             free_idx = [
                 g.index
                 for g in self.gpus
@@ -140,7 +145,7 @@ class GpuScheduler(BaseScheduler):
         with self._lock:
             self.gpus = self.inventory.refresh_usage() or self.gpus
             return {
-                "availableGpuNums": self.available_gpu_nums,
+                "availableGpuNums": self.node_gpu_count,
                 "gpuStatusMap": dict(self.gpu_status_map),
                 "gpus": [
                     {**g.to_dict(), "allocated": bool(self.gpu_status_map.get(g.uuid, 0))}

@@ -24,6 +24,7 @@ from __future__ import annotations
 
 import asyncio
 import json
+import logging
 import os
 import shutil
 import signal
@@ -37,6 +38,8 @@ from ..utils.names import safe_subpath
 from ..xerrors import ContainerExisted, ContainerNotExist, VolumeExisted
 from .base import ContainerState, GpuResolver, RuntimeDriver, VolumeState
 from .devices import visible_device_env
+
+log = logging.getLogger(__name__)
 
 CGROUP_ROOT = "/sys/fs/cgroup"
 DEFAULT_CMD = ["sleep", "infinity"]
@@ -147,7 +150,19 @@ class ProcRuntime(RuntimeDriver):
                         stderr=subprocess.DEVNULL,
                     ).returncode
                     if rc != 0:
+                        # quota silently becoming advisory is a lie to the
+                        # API — record the degradation so volume info can
+                        # surface it
                         opts["enforced"] = "none"
+                        opts["degraded"] = "remount-failed"
+                        log.warning(
+                            "volume %s: loop remount failed; size quota now advisory", name
+                        )
+                        try:
+                            with open(optf, "w") as f:
+                                json.dump(opts, f)
+                        except OSError:
+                            pass
                 self.volumes[name] = VolumeState(name=name, mountpoint=mp, options=opts)
 
     def _refresh(self, p: _Proc) -> None:
@@ -493,6 +508,11 @@ class ProcRuntime(RuntimeDriver):
                 stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
             )
             await proc.wait()
+            if os.path.ismount(vs.mountpoint):
+                # never rmtree through a still-attached mount: that deletes
+                # the volume's live data instead of the backing dir
+                self.volumes[name] = vs
+                raise RuntimeError(f"volume {name}: unmount failed, not removing")
         shutil.rmtree(vdir, ignore_errors=True)
 
     async def volume_inspect(self, name: str) -> Optional[VolumeState]:

@@ -193,7 +193,20 @@ class VolumeService:
         return out
 
     async def get_volume_info(self, name: str) -> Dict:
-        return (await self._load_spec(name)).to_dict()
+        spec = await self._load_spec(name)
+        out = spec.to_dict()
+        # surface live quota enforcement: "loop" (ENOSPC-enforced) vs
+        # "none" (advisory), plus any degradation recorded by the runtime
+        # (e.g. a failed remount after daemon restart) — a silent downgrade
+        # would misrepresent the size guarantee
+        vs = await self.runtime.volume_inspect(spec.name)
+        if vs is not None:
+            opts = vs.options or {}
+            if "enforced" in opts:
+                out["sizeEnforced"] = opts.get("enforced")
+            if opts.get("degraded"):
+                out["degraded"] = opts.get("degraded")
+        return out
 
     async def get_volume_history(self, name: str) -> List[Dict]:
         kvs = await self.store.history(Resource.VOLUMES, name)

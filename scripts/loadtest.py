@@ -66,7 +66,7 @@ async def main_async(args):
     transport = httpx.ASGITransport(app=app)
     client = httpx.AsyncClient(transport=transport, base_url="http://daemon")
 
-    n_gpus = d.gpu.available_gpu_nums
+    n_gpus = d.gpu.node_gpu_count
     stop_at = time.perf_counter() + args.seconds
     lat: list = []
     tasks = [

@@ -87,7 +87,7 @@ async def scenario1(base) -> dict:
 
 async def scenario2(base) -> dict:
     d = await fresh_daemon(base, "s2", gpus_needed=2)
-    n_gpus = d.gpu.available_gpu_nums
+    n_gpus = d.gpu.node_gpu_count
     target = 2 if n_gpus >= 2 else 0
     runs, patches = [], []
     for i in range(10):
@@ -147,7 +147,7 @@ async def scenario3(base, payload_mib: int) -> dict:
 
 async def scenario4(base) -> dict:
     d = await fresh_daemon(base, "s4")
-    n = d.gpu.available_gpu_nums
+    n = d.gpu.node_gpu_count
     t0 = time.perf_counter()
     await asyncio.gather(
         *[
@@ -172,7 +172,7 @@ async def scenario4(base) -> dict:
 
 async def scenario5(base, payload_mib: int) -> dict:
     d = await fresh_daemon(base, "s5", gpus_needed=4)
-    n_gpus = min(4, d.gpu.available_gpu_nums)
+    n_gpus = min(4, d.gpu.node_gpu_count)
     await d.replicaset.run_gpu_container(
         ContainerRun(image_name="img", replica_set_name="big", gpu_count=n_gpus)
     )

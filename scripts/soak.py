@@ -55,7 +55,7 @@ async def main_async(args):
     _mount(app, d)
     client = httpx.AsyncClient(transport=httpx.ASGITransport(app=app), base_url="http://s")
 
-    n_gpus = d.gpu.available_gpu_nums
+    n_gpus = d.gpu.node_gpu_count
     stop_at = time.perf_counter() + args.seconds
     windows: list[list[float]] = [[]]
     window_end = time.perf_counter() + args.seconds / 4
