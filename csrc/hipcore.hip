@@ -17,6 +17,7 @@
 // Build: hipcc --offload-arch=gfx950 (driven by gpu_docker_api_amd/ops/build.py).
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <map>
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
@@ -523,7 +524,67 @@ double gemm_bf16_tflops(int device, int size, int iters, int swizzle, int bk) {
   return 2.0 * size * (double)size * size / (ms * 1e9);
 }
 
-torch::Tensor gemm_bf16_8ph_bt(torch::Tensor A, torch::Tensor Bt) {
+// variant encoding: bit0 = XCD remap (T1), bits>=1 = schedule:
+// 0/1 = <XCD,PIPE=0> round-1 schedule; 2/3 = <XCD,PIPE=1> phase-ahead;
+// 4/5 = <XCD,PIPE=2> 2-tile-unrolled; 6/7 = <XCD,PIPE=3> unrolled,
+// compiler-managed lgkm waits; 8/9 = <XCD,PIPE=4> merged phases 2+3;
+// 10/11 = <XCD,PIPE=5> static-setprio younger half.
+void launch_8ph(int variant, dim3 grid, hipStream_t s,
+                const __hip_bfloat16* A, const __hip_bfloat16* Bt, float* C,
+                int M, int N, int K) {
+  switch (variant) {
+    case 1:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<1, 0>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    case 2:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<0, 1>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    case 3:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<1, 1>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    case 4:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<0, 2>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    case 5:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<1, 2>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    case 6:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<0, 3>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    case 7:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<1, 3>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    case 8:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<0, 4>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    case 9:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<1, 4>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    case 10:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<0, 5>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    case 11:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<1, 5>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+    default:
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<0, 0>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
+  }
+}
+
+torch::Tensor gemm_bf16_8ph_bt(torch::Tensor A, torch::Tensor Bt, int variant) {
   TORCH_CHECK(A.is_cuda() && Bt.is_cuda(), "GPU tensors required");
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
               Bt.scalar_type() == torch::kBFloat16, "bf16 only");
@@ -536,15 +597,14 @@ torch::Tensor gemm_bf16_8ph_bt(torch::Tensor A, torch::Tensor Bt) {
   auto C = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
   int grid = (M / 256) * (N / 256);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(gemm_bf16_8ph::gemm_bf16_8phase_kernel, dim3(grid),
-                     dim3(512), 0, stream.stream(),
-                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
-                     reinterpret_cast<const __hip_bfloat16*>(Bt.data_ptr()),
-                     C.data_ptr<float>(), M, N, K);
+  launch_8ph(variant, dim3(grid), stream.stream(),
+             reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
+             reinterpret_cast<const __hip_bfloat16*>(Bt.data_ptr()),
+             C.data_ptr<float>(), M, N, K);
   return C;
 }
 
-double gemm_bf16_8ph_tflops(int device, int size, int iters) {
+double gemm_bf16_8ph_tflops(int device, int size, int iters, int variant) {
   HIP_CHECK(hipSetDevice(device));
   mfma_warmup(device, 20000);
   size_t n = (size_t)size * size;
@@ -560,14 +620,50 @@ double gemm_bf16_8ph_tflops(int device, int size, int iters) {
   double ms = time_kernel_ms(
       device,
       [&](hipStream_t s) {
-        hipLaunchKernelGGL(gemm_bf16_8ph::gemm_bf16_8phase_kernel, dim3(grid),
-                           dim3(512), 0, s, A, Bt, C, size, size, size);
+        launch_8ph(variant, dim3(grid), s, A, Bt, C, size, size, size);
       },
       iters);
   (void)hipFree(A);
   (void)hipFree(Bt);
   (void)hipFree(C);
   return 2.0 * size * (double)size * size / (ms * 1e9);
+}
+
+// Within-probe interleaved A/B over the template variants (guide §5.4
+// rules 9/24: run-to-run noise ~±3%, so variants must be interleaved in one
+// process). Returns {variant: [tflops per round]}.
+py::dict gemm_bf16_8ph_ab(int device, int size, int iters, int rounds,
+                          std::vector<int> variants) {
+  HIP_CHECK(hipSetDevice(device));
+  mfma_warmup(device, 20000);
+  size_t n = (size_t)size * size;
+  __hip_bfloat16 *A = nullptr, *Bt = nullptr;
+  float* C = nullptr;
+  HIP_CHECK(hipMalloc(&A, n * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&Bt, n * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&C, n * sizeof(float)));
+  hipLaunchKernelGGL(fill_bf16_hash_kernel, dim3(4096), dim3(256), 0, 0, A, n, 1u);
+  hipLaunchKernelGGL(fill_bf16_hash_kernel, dim3(4096), dim3(256), 0, 0, Bt, n, 7u);
+  HIP_CHECK(hipDeviceSynchronize());
+  int grid = (size / 256) * (size / 256);
+  std::map<int, std::vector<double>> out;
+  for (int r = 0; r < rounds; ++r) {
+    for (int v : variants) {
+      double ms = time_kernel_ms(
+          device,
+          [&](hipStream_t s) {
+            launch_8ph(v, dim3(grid), s, A, Bt, C, size, size, size);
+          },
+          iters);
+      out[v].push_back(2.0 * size * (double)size * size / (ms * 1e9));
+    }
+  }
+  (void)hipFree(A);
+  (void)hipFree(Bt);
+  (void)hipFree(C);
+  py::dict d;
+  for (auto& kv : out) d[py::int_(kv.first)] = kv.second;
+  return d;
 }
 
 int device_count() {
@@ -606,9 +702,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_warmup", &mfma_warmup, py::arg("device") = 0,
         py::arg("spins") = 20000);
   m.def("gemm_bf16_bt", &gemm_bf16_bt, py::arg("A"), py::arg("Bt"));
-  m.def("gemm_bf16_8ph", &gemm_bf16_8ph_bt, py::arg("A"), py::arg("Bt"));
+  m.def("gemm_bf16_8ph", &gemm_bf16_8ph_bt, py::arg("A"), py::arg("Bt"),
+        py::arg("variant") = 0);
+  m.def("gemm_bf16_8ph_ab", &gemm_bf16_8ph_ab, py::arg("device") = 0,
+        py::arg("size") = 4096, py::arg("iters") = 4, py::arg("rounds") = 3,
+        py::arg("variants") = std::vector<int>{0, 1, 2, 3});
   m.def("gemm_bf16_8ph_tflops", &gemm_bf16_8ph_tflops, py::arg("device") = 0,
-        py::arg("size") = 4096, py::arg("iters") = 10,
+        py::arg("size") = 4096, py::arg("iters") = 10, py::arg("variant") = 0,
         py::call_guard<py::gil_scoped_release>());
   m.def("gemm_bf16_tflops", &gemm_bf16_tflops, py::arg("device") = 0,
         py::arg("size") = 4096, py::arg("iters") = 10, py::arg("swizzle") = 0,
