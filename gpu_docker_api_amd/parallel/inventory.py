@@ -269,6 +269,11 @@ class MockInventory(GpuInventory):
                 uuid=f"MockMI355X-{i}",
                 bdf=f"0000:{0x10 + i:02x}:00.0",
                 vram_used=self._used.get(i, 0),
+                # synthetic DRM nodes (renderD128+i mirrors amdgpu's
+                # numbering) so the docker device-injection path is
+                # exercised by the mock backend too
+                render_node=f"/dev/dri/renderD{128 + i}",
+                card_node=f"/dev/dri/card{i}",
             )
             for i in range(self.count)
         ]

@@ -81,6 +81,20 @@ def build_fake_etcd(store: MemoryMVCC | None = None) -> FastAPI:
             "count": str(len(kvs)),
         }
 
+    @app.post("/v3/kv/compaction")
+    async def compaction(request: Request):
+        body = await request.json()
+        rev = int(body.get("revision", 0) or 0)
+        try:
+            mvcc.compact(rev)
+        except RevisionCompacted:
+            return JSONResponse(
+                status_code=400,
+                content={"code": 11,
+                         "message": "etcdserver: mvcc: required revision has been compacted"},
+            )
+        return {"header": header()}
+
     @app.post("/v3/kv/deleterange")
     async def deleterange(request: Request):
         body = await request.json()
@@ -92,3 +106,21 @@ def build_fake_etcd(store: MemoryMVCC | None = None) -> FastAPI:
         return {"header": header(), "deleted": str(deleted)}
 
     return app
+
+
+def main() -> None:
+    """Serve the gateway on TCP for out-of-process integration tests / dev:
+    ``python -m gpu_docker_api_amd.state.etcd_fake --port 2379``."""
+    import argparse
+
+    import uvicorn
+
+    p = argparse.ArgumentParser()
+    p.add_argument("--host", default="127.0.0.1")
+    p.add_argument("--port", type=int, required=True)
+    args = p.parse_args()
+    uvicorn.run(build_fake_etcd(), host=args.host, port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()
