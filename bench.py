@@ -64,6 +64,31 @@ class DaemonThread:
     def __init__(self, port: int, n_gpus: int, use_gpu: bool, data_dir: str, runtime: str = "proc"):
         from gpu_docker_api_amd.config import Config
 
+        self.engine = "dockerd"
+        self._engine_proc = None
+        docker_socket = "/var/run/docker.sock"
+        if runtime == "docker" and not os.path.exists(docker_socket):
+            # no real dockerd in this environment: stand up the Engine-API
+            # simulator (ProcRuntime-backed) on a private socket so the
+            # docker code path is still the one measured; recorded in the
+            # output JSON as engine=dockerd-sim
+            import subprocess
+            import sys as _sys
+
+            docker_socket = os.path.join(data_dir, "dockerd.sock")
+            os.makedirs(data_dir, exist_ok=True)
+            self._engine_proc = subprocess.Popen(
+                [_sys.executable, "-m", "gpu_docker_api_amd.testing.dockerd_sim",
+                 "--socket", docker_socket, "--data", os.path.join(data_dir, "engine")],
+                stdout=subprocess.DEVNULL,
+                stderr=subprocess.DEVNULL,
+                start_new_session=True,
+            )
+            deadline = time.time() + 30
+            while time.time() < deadline and not os.path.exists(docker_socket):
+                time.sleep(0.05)
+            self.engine = "dockerd-sim"
+
         self.cfg = Config(
             addr=f"127.0.0.1:{port}",
             state="memory",
@@ -74,6 +99,7 @@ class DaemonThread:
             copy_engine="auto",
             run_xgmi_probe=use_gpu,  # native HIP probe, outside the timed region
             port_range="41000-42000",
+            docker_socket=docker_socket,
         )
         self.port = port
         self._thread: threading.Thread | None = None
@@ -106,6 +132,12 @@ class DaemonThread:
             self._server.should_exit = True
         if self._thread is not None:
             self._thread.join(timeout=30)
+        if self._engine_proc is not None:
+            self._engine_proc.terminate()
+            try:
+                self._engine_proc.wait(timeout=10)
+            except Exception:
+                self._engine_proc.kill()
 
 
 def one_cycle(client, name: str, patch_to: int) -> float:
@@ -160,16 +192,17 @@ def main():
     port = daemon_port(args)
     server = None
     if rank == 0:
+        import shutil
+
+        data_dir = os.path.join("/tmp", f"gda-bench-{port}")
+        shutil.rmtree(data_dir, ignore_errors=True)
         server = DaemonThread(
             port,
             n_gpus,
             use_gpu,
-            data_dir=os.path.join("/tmp", f"gda-bench-{port}"),
+            data_dir=data_dir,
             runtime=args.runtime,
         )
-        import shutil
-
-        shutil.rmtree(server.cfg.data_dir, ignore_errors=True)
         server.start()
     barrier_sync()
 
@@ -238,6 +271,7 @@ def main():
                 "seq_len": 0,
                 "parallelism": f"{world} concurrent tenants, 1 daemon, {n_gpus} GPUs",
                 "runtime": args.runtime,
+                "engine": (server.engine if server is not None and args.runtime == "docker" else None),
                 "patch": f"gpuCount 1->{patch_to}",
                 "deviation": deviation,
                 "throughput_cycles_per_s": round(world * args.steps / max_elapsed, 3),
