@@ -92,10 +92,52 @@ class CopyEngine:
     def __init__(self, kind: str = "auto") -> None:
         self.kind = kind
 
+    # trees at or below this size are copied inline: engine setup (io_uring
+    # ring init / tar fork+exec / thread hop) costs ~1-2 ms, which dominates
+    # the rolling-replace latency when the writable layer is near-empty
+    SMALL_TREE_BYTES = 1 << 20
+    SMALL_TREE_FILES = 32
+
+    @classmethod
+    def _small_tree_entries(cls, src: str):
+        """Top-level regular files/symlinks totaling <= SMALL_TREE_BYTES,
+        or None when the tree needs a real engine (subdirs/large/special)."""
+        entries = []
+        total = 0
+        try:
+            with os.scandir(src) as it:
+                for e in it:
+                    if len(entries) >= cls.SMALL_TREE_FILES:
+                        return None
+                    if e.is_symlink():
+                        entries.append(e)
+                        continue
+                    if not e.is_file(follow_symlinks=False):
+                        return None
+                    total += e.stat(follow_symlinks=False).st_size
+                    if total > cls.SMALL_TREE_BYTES:
+                        return None
+                    entries.append(e)
+        except OSError:
+            return None
+        return entries
+
     async def copy_dir(self, src: str, dest: str) -> None:
         if not os.path.isdir(src):
             raise FileNotFoundError(src)
         os.makedirs(dest, exist_ok=True)
+        small = self._small_tree_entries(src)
+        if small is not None:
+            for e in small:
+                target = os.path.join(dest, e.name)
+                if e.is_symlink():
+                    link = os.readlink(e.path)
+                    if os.path.lexists(target):
+                        os.unlink(target)
+                    os.symlink(link, target)
+                else:
+                    shutil.copy2(e.path, target)
+            return
         kind = self.kind
         if kind == "auto":
             kind = "iouring" if _load_iocopy() is not None else "tar"
