@@ -1,0 +1,259 @@
+// MX-fp8 (OCP e4m3) MFMA GEMM — 256^2 tile, 8-phase schedule, K-step 128.
+//
+// gfx950's only high-rate fp8 path is the block-scaled MX instruction
+// (guide §3: there is NO non-scaled mfma_f32_*x64_fp8; the MX-scaled
+// 16x16x128 runs at ~4.6 PF µbench vs 2.1 PF for non-scaled fp8). This
+// kernel runs it as a plain fp8 GEMM: every 32-element block scale is
+// e8m0 127 (x1.0), so numerics are exactly "e4m3 inputs, fp32 accumulate".
+//
+// Operand layout (verified on hardware by csrc-probe, max err 0.0 vs exact
+// integer reference — see profiles/gemm_kernel_stats.md round-2 fp8 note):
+//   A: lane l holds row (l&15), k bytes (l>>4)*32 .. +32 (8 VGPRs, v8i32)
+//   B: lane l holds col (l&15) of B = row of Bt, same k range
+//   C/D: col = lane&15, row = (lane>>4)*4 + reg (shape-determined,
+//        dtype-independent on gfx950)
+//   scales: opsel 0, low byte 127 on both operands; cbsz=0 blgp=0 (e4m3).
+//
+// Structure: the bf16 8-phase template's schedule verbatim (PIPE=0 of
+// gemm_bf16_8phase.hip — measured optimal there across 6 variant families)
+// with BK=128 fp8: a half-tile is [128][128] fp8 = 16 KiB — the SAME byte
+// shape as the bf16 [128][64] half, so staging (2 glds x 16B per thread),
+// the chunk-xor LDS swizzle (byte ^= (row&7)<<4), slot cadence, counted
+// vmcnt and barriers are unchanged; only the fragment reads (2x
+// ds_read_b128 = 32 B per fragment) and the MFMA (8 independent
+// 16x16x128 per phase, no kk loop) differ.
+//
+// Constraints: M,N multiples of 256; K multiple of 128, K >= 256.
+#include <hip/hip_runtime.h>
+
+namespace gemm_fp8_mx {
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using i32x4 = __attribute__((ext_vector_type(4))) int;
+using i32x8 = __attribute__((ext_vector_type(8))) int;
+
+constexpr int BM = 256;
+constexpr int BN = 256;
+constexpr int BK = 128;                  // fp8 K-step of the MX instruction
+constexpr int THREADS = 512;             // 8 waves, 2(M) x 4(N)
+constexpr int HALF_BYTES = 128 * BK;     // [128][128] fp8 = 16 KiB
+constexpr int SLOTS = 8;
+
+__device__ inline void glds16(const unsigned char* gsrc, unsigned char* lds) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)gsrc,
+      (__attribute__((address_space(3))) unsigned int*)lds, 16, 0, 0);
+}
+
+// chunk-xor swizzle in 16-B granules (byte ^= (row&7)<<4) — the form that
+// measured conflict-free on the bf16 sibling (same 128-B row stride)
+__device__ inline int swz_byte(int row, int byte_off) {
+  return byte_off ^ ((row & 7) << 4);
+}
+
+__device__ inline void stage_half(const unsigned char* gbase, int ldk,
+                                  unsigned char* half_base) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    int slot = pass * THREADS + t;
+    int row_l = slot >> 3;
+    int byte_off = (slot & 7) * 16;
+    glds16(gbase + row_l * ldk + swz_byte(row_l, byte_off),
+           half_base + row_l * BK + byte_off);
+  }
+}
+
+// one 32-B MFMA fragment: two independently-swizzled 16-B chunks
+__device__ inline i32x8 read_frag(const unsigned char* half_base, int row,
+                                  int kb) {
+  i32x4 lo = *reinterpret_cast<const i32x4*>(half_base + row * BK +
+                                             swz_byte(row, kb));
+  i32x4 hi = *reinterpret_cast<const i32x4*>(half_base + row * BK +
+                                             swz_byte(row, kb + 16));
+  return i32x8{lo[0], lo[1], lo[2], lo[3], hi[0], hi[1], hi[2], hi[3]};
+}
+
+__device__ inline void wait_lgkm0_fence() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_sched_barrier(0);
+}
+
+__device__ inline void wait_vmcnt(int halves_outstanding) {
+  switch (halves_outstanding) {
+    case 0: asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); break;
+    case 1: asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); break;
+    case 2: asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); break;
+    default: asm volatile("s_waitcnt vmcnt(6)" ::: "memory"); break;
+  }
+  __builtin_amdgcn_sched_barrier(0);
+}
+
+#define MX_MFMA(a, b, c) \
+  __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4((a), (b), (c), 0, 0, 0, 127, 0, 127)
+
+__global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
+    const unsigned char* __restrict__ A,   // [M][K] e4m3
+    const unsigned char* __restrict__ Bt,  // [N][K] e4m3
+    float* __restrict__ C,                 // [M][N]
+    int M, int N, int K) {
+  __shared__ unsigned char lds[SLOTS * HALF_BYTES];  // 128 KiB
+
+  const int tiles_n = N / BN;
+  const int bm = ((int)blockIdx.x / tiles_n) * BM;
+  const int bn = ((int)blockIdx.x % tiles_n) * BN;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+
+  const int frow = lane & 15;
+  const int fkb = (lane >> 4) * 32;  // 32-byte k-slice per lane
+
+  const unsigned char* Ablk = A + (long)bm * K;
+  const unsigned char* Bblk = Bt + (long)bn * K;
+  const int T = K / BK;
+
+  f32x4 acc[2][4][2][2] = {};
+
+  auto stage_h = [&](int h) {
+    int c = h & 3;
+    int tt = h >> 2;
+    const unsigned char* g;
+    if (c == 0)
+      g = Ablk + tt * BK;
+    else if (c == 1)
+      g = Bblk + tt * BK;
+    else if (c == 2)
+      g = Ablk + (long)128 * K + tt * BK;
+    else
+      g = Bblk + (long)128 * K + tt * BK;
+    stage_half(g, K, lds + (h & 7) * HALF_BYTES);
+  };
+
+  for (int h = 0; h < 4 && h < 4 * T; ++h) stage_h(h);
+  wait_vmcnt(2);
+  for (int h = 4; h < 7 && h < 4 * T; ++h) stage_h(h);
+  wait_vmcnt(3);
+  __builtin_amdgcn_s_barrier();
+
+  i32x8 afrag[4];
+  i32x8 bfrag[2][2];
+
+  const int arow = wm * 64 + frow;
+  const int brow = wn * 32 + frow;
+
+  for (int t = 0; t < T; ++t) {
+    const unsigned char* As0 = lds + ((4 * t + 0) & 7) * HALF_BYTES;
+    const unsigned char* Bs0 = lds + ((4 * t + 1) & 7) * HALF_BYTES;
+    const unsigned char* As1 = lds + ((4 * t + 2) & 7) * HALF_BYTES;
+    const unsigned char* Bs1 = lds + ((4 * t + 3) & 7) * HALF_BYTES;
+
+    // ---- phase 0: q(0,0); read A(qm0)+B set 0; stage 4t+7
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+      afrag[fm] = read_frag(As0, arow + fm * 16, fkb);
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+      bfrag[0][fn] = read_frag(Bs0, brow + fn * 16, fkb);
+    if (4 * t + 7 < 4 * T) stage_h(4 * t + 7);
+    asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    wait_lgkm0_fence();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+        acc[0][fm][0][fn] = MX_MFMA(afrag[fm], bfrag[0][fn], acc[0][fm][0][fn]);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+
+    // ---- phase 1: q(0,1); read B set 1; stage 4t+8
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+      bfrag[1][fn] = read_frag(Bs1, brow + fn * 16, fkb);
+    if (4 * t + 8 < 4 * T) stage_h(4 * t + 8);
+    __builtin_amdgcn_s_barrier();
+    wait_lgkm0_fence();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+        acc[0][fm][1][fn] = MX_MFMA(afrag[fm], bfrag[1][fn], acc[0][fm][1][fn]);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+
+    // ---- phase 2: q(1,1); re-read A(qm1); stage 4t+9
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+      afrag[fm] = read_frag(As1, arow + fm * 16, fkb);
+    if (4 * t + 9 < 4 * T) stage_h(4 * t + 9);
+    __builtin_amdgcn_s_barrier();
+    wait_lgkm0_fence();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+        acc[1][fm][1][fn] = MX_MFMA(afrag[fm], bfrag[1][fn], acc[1][fm][1][fn]);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+
+    // ---- phase 3: q(1,0); B set 0 still live; stage 4t+10
+    if (4 * t + 10 < 4 * T) stage_h(4 * t + 10);
+    __builtin_amdgcn_s_barrier();
+    wait_lgkm0_fence();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+        acc[1][fm][0][fn] = MX_MFMA(afrag[fm], bfrag[0][fn], acc[1][fm][0][fn]);
+    __builtin_amdgcn_s_setprio(0);
+    {
+      int staged = min(4 * T, 4 * t + 11);
+      int allowed = staged - 4 * (t + 2);
+      wait_vmcnt(allowed < 0 ? 0 : (allowed > 3 ? 3 : allowed));
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  const int ccol = lane & 15;
+  const int crow = (lane >> 4) * 4;
+#pragma unroll
+  for (int qm = 0; qm < 2; ++qm)
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int qn = 0; qn < 2; ++qn)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          int row0 = bm + qm * 128 + wm * 64 + fm * 16 + crow;
+          int col = bn + qn * 128 + wn * 32 + fn * 16 + ccol;
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            C[(long)(row0 + r) * N + col] = acc[qm][fm][qn][fn][r];
+        }
+}
+
+// pseudorandom VALID e4m3 fill: full sign/mantissa variation, exponents
+// bounded so products stay finite (bench-honesty: zero or sign-stuck fills
+// inflate TF via DVFS — guide §5.4 rule 25)
+__global__ void fill_e4m3_hash_kernel(unsigned char* p, size_t n, unsigned seed) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    unsigned h = (unsigned)(i * 2654435761u) ^ seed;
+    h ^= h >> 13;
+    h *= 0x85ebca6bu;
+    h ^= h >> 16;
+    // sign = bit 0; exponent in [4..11] (values ~2^-3 .. 2^4); mantissa 3 bits
+    unsigned char b = (unsigned char)(((h & 1u) << 7) | ((((h >> 1) & 7u) + 4u) << 3) |
+                                      ((h >> 4) & 7u));
+    p[i] = b;
+  }
+}
+
+}  // namespace gemm_fp8_mx

@@ -188,6 +188,7 @@ __global__ void mfma_warmup_kernel(float* __restrict__ sink, int iters) {
 
 #include "gemm_bf16.hip"
 #include "gemm_bf16_8phase.hip"
+#include "gemm_fp8_mx.hip"
 
 namespace {
 
@@ -666,6 +667,56 @@ py::dict gemm_bf16_8ph_ab(int device, int size, int iters, int rounds,
   return d;
 }
 
+torch::Tensor gemm_fp8_mx_bt(torch::Tensor A, torch::Tensor Bt) {
+  TORCH_CHECK(A.is_cuda() && Bt.is_cuda(), "GPU tensors required");
+  TORCH_CHECK(A.scalar_type() == torch::kUInt8 &&
+              Bt.scalar_type() == torch::kUInt8,
+              "raw e4m3 bytes expected (uint8 view of float8_e4m3fn)");
+  A = A.contiguous();
+  Bt = Bt.contiguous();
+  int M = A.size(0), K = A.size(1), N = Bt.size(0);
+  TORCH_CHECK(Bt.size(1) == K, "Bt must be [N][K]");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 128 == 0 && K >= 256,
+              "M,N multiples of 256; K multiple of 128, >= 256");
+  auto C = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
+  int grid = (M / 256) * (N / 256);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel, dim3(grid), dim3(512),
+                     0, stream.stream(),
+                     reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                     reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                     C.data_ptr<float>(), M, N, K);
+  return C;
+}
+
+double gemm_fp8_mx_tflops(int device, int size, int iters) {
+  HIP_CHECK(hipSetDevice(device));
+  mfma_warmup(device, 20000);
+  size_t n = (size_t)size * size;
+  unsigned char *A = nullptr, *Bt = nullptr;
+  float* C = nullptr;
+  HIP_CHECK(hipMalloc(&A, n));
+  HIP_CHECK(hipMalloc(&Bt, n));
+  HIP_CHECK(hipMalloc(&C, n * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fp8_mx::fill_e4m3_hash_kernel, dim3(4096), dim3(256),
+                     0, 0, A, n, 1u);
+  hipLaunchKernelGGL(gemm_fp8_mx::fill_e4m3_hash_kernel, dim3(4096), dim3(256),
+                     0, 0, Bt, n, 7u);
+  HIP_CHECK(hipDeviceSynchronize());
+  int grid = (size / 256) * (size / 256);
+  double ms = time_kernel_ms(
+      device,
+      [&](hipStream_t s) {
+        hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel, dim3(grid),
+                           dim3(512), 0, s, A, Bt, C, size, size, size);
+      },
+      iters);
+  (void)hipFree(A);
+  (void)hipFree(Bt);
+  (void)hipFree(C);
+  return 2.0 * size * (double)size * size / (ms * 1e9);
+}
+
 int device_count() {
   int n = 0;
   if (hipGetDeviceCount(&n) != hipSuccess) return 0;
@@ -704,6 +755,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16_bt", &gemm_bf16_bt, py::arg("A"), py::arg("Bt"));
   m.def("gemm_bf16_8ph", &gemm_bf16_8ph_bt, py::arg("A"), py::arg("Bt"),
         py::arg("variant") = 0);
+  m.def("gemm_fp8_mx", &gemm_fp8_mx_bt, py::arg("A"), py::arg("Bt"));
+  m.def("gemm_fp8_mx_tflops", &gemm_fp8_mx_tflops, py::arg("device") = 0,
+        py::arg("size") = 4096, py::arg("iters") = 10,
+        py::call_guard<py::gil_scoped_release>());
   m.def("gemm_bf16_8ph_ab", &gemm_bf16_8ph_ab, py::arg("device") = 0,
         py::arg("size") = 4096, py::arg("iters") = 4, py::arg("rounds") = 3,
         py::arg("variants") = std::vector<int>{0, 1, 2, 3});

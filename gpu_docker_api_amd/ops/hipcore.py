@@ -88,10 +88,15 @@ async def run_probe_async(mib: int = 256, iters: int = 5) -> dict:
     )
 
 
-def validate_gpus(indices: Optional[list] = None, size: int = 4096, iters: int = 5) -> dict:
+def validate_gpus(
+    indices: Optional[list] = None, size: int = 4096, iters: int = 5, fp8: bool = True
+) -> dict:
     """Per-GPU health burn-in: HBM stream bandwidth + dense bf16 MFMA GEMM
-    throughput (guide's 128^2-tile structure). Run on FREE GPUs before
-    scheduling; a GPU far below its siblings is flagged."""
+    (the 8-phase 256^2 structure, ~1.1 PF) + the MX-fp8 path (block-scaled
+    mfma_scale 16x16x128, ~1.9 PF — exercised because fp8 training is what
+    tenants run on MI355X and its datapath can fail independently of bf16).
+    Run on FREE GPUs before scheduling; a GPU far below its siblings is
+    flagged."""
     ext = load_ext()
     n = ext.device_count()
     idx = list(indices) if indices else list(range(n))
@@ -107,13 +112,14 @@ def validate_gpus(indices: Optional[list] = None, size: int = 4096, iters: int =
             tflops = ext.gemm_bf16_8ph_tflops(i, size, iters)
         else:
             tflops = ext.gemm_bf16_tflops(i, size, iters)
-        report["gpus"].append(
-            {
-                "index": i,
-                "hbm_gbps": round(hbm, 1),
-                "bf16_tflops": round(tflops, 1),
-            }
-        )
+        entry = {
+            "index": i,
+            "hbm_gbps": round(hbm, 1),
+            "bf16_tflops": round(tflops, 1),
+        }
+        if fp8 and size % 256 == 0 and size >= 256:
+            entry["fp8_tflops"] = round(ext.gemm_fp8_mx_tflops(i, size, iters), 1)
+        report["gpus"].append(entry)
     vals = [g["bf16_tflops"] for g in report["gpus"]]
     if vals:
         top = max(vals)

@@ -1,0 +1,39 @@
+#!/usr/bin/env python3
+"""MX-fp8 GEMM numerics race-screen + throughput on a real MI355X."""
+import json
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def main():
+    import torch
+
+    from gpu_docker_api_amd.ops import hipcore
+
+    ext = hipcore.load_ext()
+    torch.manual_seed(9)
+    out = {}
+    for (M, N, K) in ((256, 256, 256), (512, 256, 512), (4096, 4096, 4096)):
+        A = (torch.randn(M, K, device="cuda") * 0.5).to(torch.float8_e4m3fn)
+        Bt = (torch.randn(N, K, device="cuda") * 0.5).to(torch.float8_e4m3fn)
+        ref = A.float() @ Bt.float().T
+        scale = ref.abs().max().item() + 1e-6
+        worst = 0.0
+        for _ in range(4):
+            C = ext.gemm_fp8_mx(A.view(torch.uint8), Bt.view(torch.uint8))
+            torch.cuda.synchronize()
+            worst = max(worst, (C - ref).abs().max().item() / scale)
+        out[f"relerr_{M}x{N}x{K}"] = worst
+    for size, iters in ((4096, 8), (8192, 4)):
+        out[f"tflops_{size}"] = [
+            round(ext.gemm_fp8_mx_tflops(0, size, iters), 1) for _ in range(3)
+        ]
+    print(json.dumps(out))
+    os.makedirs("gpurun_out", exist_ok=True)
+    open("gpurun_out/fp8_mx.json", "w").write(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

@@ -308,3 +308,24 @@ def test_bench_short_run():
     assert result["n_gpus"] == 1
     assert result["value"] > 0
     assert result["higher_is_better"] is False
+
+
+def test_gemm_fp8_mx_numerics_and_throughput(ext):
+    """MX-fp8 (block-scaled mfma 16x16x128, scales pinned to 1.0): inputs
+    are exact e4m3, so vs the dequantized fp32 torch reference only
+    accumulation-order noise remains. Measured round 2: 1666-1746 TF
+    @4096^3 on random operands (profiles/gemm_fp8_mx.json)."""
+    import torch
+
+    torch.manual_seed(9)
+    A = (torch.randn(512, 512, device="cuda") * 0.5).to(torch.float8_e4m3fn)
+    Bt = (torch.randn(256, 512, device="cuda") * 0.5).to(torch.float8_e4m3fn)
+    ref = A.float() @ Bt.float().T
+    scale = ref.abs().max().item() + 1e-6
+    for _ in range(3):  # race screen: nondeterminism would betray a pipeline bug
+        C = ext.gemm_fp8_mx(A.view(torch.uint8), Bt.view(torch.uint8))
+        err = (C - ref).abs().max().item() / scale
+        assert err < 1e-3, f"fp8 MX rel err {err}"
+    tflops = ext.gemm_fp8_mx_tflops(0, 4096, 8)
+    print(f"fp8 MX GEMM (8-phase 256^2, K=128): {tflops:.0f} TFLOPS @4096^3")
+    assert tflops > 1200, f"fp8 MX GEMM regressed: {tflops} TF (floor 1200 = 75% of measured 1666)"
