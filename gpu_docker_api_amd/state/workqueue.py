@@ -44,6 +44,10 @@ class WorkQueue:
         self._q: asyncio.Queue = asyncio.Queue(maxsize=maxsize)
         self._task: asyncio.Task | None = None
         self._closing = False
+        # items sitting in a retry backoff: neither in the queue nor done,
+        # so drain() must wait for them too (an etcd outage would otherwise
+        # let shutdown proceed while writes are still pending)
+        self._retrying = 0
 
     def start(self) -> None:
         if self._task is None:
@@ -90,15 +94,26 @@ class WorkQueue:
                         delay = min(0.05 * (2 ** item.attempts), 5.0)
                         log.warning("workqueue: retrying %r in %.2fs: %s", item, delay, exc)
                         loop = asyncio.get_running_loop()
-                        loop.call_later(delay, self.enqueue, item)
+                        self._retrying += 1
+                        loop.call_later(delay, self._requeue, item)
             finally:
                 self._q.task_done()
 
+    def _requeue(self, item) -> None:
+        self._retrying -= 1
+        self.enqueue(item)
+
     async def drain(self) -> None:
-        """Wait until everything currently enqueued is flushed."""
-        await self._q.join()
+        """Wait until everything enqueued OR in a retry backoff is flushed
+        (or dropped at max_attempts)."""
+        while True:
+            await self._q.join()
+            if self._retrying == 0 and self._q.empty():
+                return
+            await asyncio.sleep(0.05)
 
     async def close(self) -> None:
+        await self.drain()
         self._closing = True
         await self._q.join()
         if self._task is not None:

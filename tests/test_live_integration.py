@@ -307,3 +307,73 @@ def test_daemon_full_flow_on_live_etcd(etcd_live, tmp_path, run):
         await d2.stop()
 
     run(main())
+
+
+def test_workqueue_survives_etcd_outage(tmp_path, run):
+    """Kill the live etcd process mid-traffic; the write-behind queue must
+    retry through the outage and flush once a replacement serves the same
+    port (the reference retries forever but its shutdown can drop queued
+    writes; here drain() also waits out retry backoffs)."""
+    import httpx
+
+    from gpu_docker_api_amd.state.etcd_gateway import EtcdGatewayStore
+    from gpu_docker_api_amd.state.keys import Resource
+    from gpu_docker_api_amd.state.workqueue import WorkQueue
+
+    port = _free_port()
+
+    def spawn():
+        return subprocess.Popen(
+            [sys.executable, "-m", "gpu_docker_api_amd.state.etcd_fake",
+             "--port", str(port)],
+            cwd=REPO,
+            stdout=subprocess.DEVNULL,
+            stderr=subprocess.DEVNULL,
+            start_new_session=True,
+        )
+
+    def up() -> bool:
+        try:
+            with socket.create_connection(("127.0.0.1", port), timeout=0.2):
+                return True
+        except OSError:
+            return False
+
+    proc = spawn()
+    replacement = None
+    try:
+        _wait_for(up, what="etcd gateway port")
+
+        async def main():
+            nonlocal proc, replacement
+            store = EtcdGatewayStore(f"http://127.0.0.1:{port}", timeout=0.5)
+            q = WorkQueue(store)
+            q.start()
+            q.put(Resource.GPUS, "gpuStatusMapKey", "before-outage")
+            await q.drain()
+
+            # hard-kill etcd, then enqueue during the outage
+            proc.kill()
+            proc.wait(timeout=10)
+            q.put(Resource.GPUS, "gpuStatusMapKey", "during-outage")
+            await asyncio.sleep(0.3)  # let the first attempts fail
+
+            replacement = spawn()
+            deadline = time.time() + 20
+            while time.time() < deadline and not up():
+                await asyncio.sleep(0.05)
+            await q.drain()
+            kv = await store.get(Resource.GPUS, "gpuStatusMapKey")
+            assert kv.value == "during-outage"
+            await q.close()
+            await store.close()
+
+        run(main())
+    finally:
+        for p in (proc, replacement):
+            if p is not None:
+                try:
+                    p.kill()
+                    p.wait(timeout=5)
+                except Exception:
+                    pass
