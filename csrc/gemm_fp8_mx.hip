@@ -23,7 +23,16 @@
 // ds_read_b128 = 32 B per fragment) and the MFMA (8 independent
 // 16x16x128 per phase, no kk loop) differ.
 //
-// Constraints: M,N multiples of 256; K multiple of 128, K >= 256.
+// FMT template axis: 0 = fp8 e4m3 (K-step 128), 4 = fp4 e2m1 (K-step 256 —
+// nibble-packed, so a [128][256-fp4] half-tile is the SAME 16 KiB byte
+// shape; each quadrant phase then runs two 16x16x128 MFMAs per fragment
+// pair, one per 128-element k-half, at the 4x fp4 rate). The fp4 operand
+// layout (low nibble = even k; 16 bytes in the low 4 dwords of the v8i32
+// operand; cbsz=blgp=4) was verified on hardware with the same
+// single-MFMA exact-integer probe as fp8 (max abs err 0.0).
+//
+// Constraints: M,N multiples of 256; K multiple of 128 (fp8) / 256 (fp4),
+// K >= 2 K-steps.
 #include <hip/hip_runtime.h>
 
 namespace gemm_fp8_mx {
@@ -34,9 +43,9 @@ using i32x8 = __attribute__((ext_vector_type(8))) int;
 
 constexpr int BM = 256;
 constexpr int BN = 256;
-constexpr int BK = 128;                  // fp8 K-step of the MX instruction
+constexpr int BKB = 128;                 // K-step in BYTES (fp8: 128 elems; fp4: 256)
 constexpr int THREADS = 512;             // 8 waves, 2(M) x 4(N)
-constexpr int HALF_BYTES = 128 * BK;     // [128][128] fp8 = 16 KiB
+constexpr int HALF_BYTES = 128 * BKB;    // [128][128 B] = 16 KiB
 constexpr int SLOTS = 8;
 
 __device__ inline void glds16(const unsigned char* gsrc, unsigned char* lds) {
@@ -60,18 +69,26 @@ __device__ inline void stage_half(const unsigned char* gbase, int ldk,
     int row_l = slot >> 3;
     int byte_off = (slot & 7) * 16;
     glds16(gbase + row_l * ldk + swz_byte(row_l, byte_off),
-           half_base + row_l * BK + byte_off);
+           half_base + row_l * BKB + byte_off);
   }
 }
 
-// one 32-B MFMA fragment: two independently-swizzled 16-B chunks
+// one fp8 MFMA fragment (32 B): two independently-swizzled 16-B chunks
 __device__ inline i32x8 read_frag(const unsigned char* half_base, int row,
                                   int kb) {
-  i32x4 lo = *reinterpret_cast<const i32x4*>(half_base + row * BK +
+  i32x4 lo = *reinterpret_cast<const i32x4*>(half_base + row * BKB +
                                              swz_byte(row, kb));
-  i32x4 hi = *reinterpret_cast<const i32x4*>(half_base + row * BK +
+  i32x4 hi = *reinterpret_cast<const i32x4*>(half_base + row * BKB +
                                              swz_byte(row, kb + 16));
   return i32x8{lo[0], lo[1], lo[2], lo[3], hi[0], hi[1], hi[2], hi[3]};
+}
+
+// one fp4 MFMA fragment (16 B = 32 nibbles) in the low 4 dwords
+__device__ inline i32x8 read_frag4(const unsigned char* half_base, int row,
+                                   int kb) {
+  i32x4 lo = *reinterpret_cast<const i32x4*>(half_base + row * BKB +
+                                             swz_byte(row, kb));
+  return i32x8{lo[0], lo[1], lo[2], lo[3], 0, 0, 0, 0};
 }
 
 __device__ inline void wait_lgkm0_fence() {
@@ -89,14 +106,15 @@ __device__ inline void wait_vmcnt(int halves_outstanding) {
   __builtin_amdgcn_sched_barrier(0);
 }
 
-#define MX_MFMA(a, b, c) \
-  __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4((a), (b), (c), 0, 0, 0, 127, 0, 127)
+#define MX_MFMA(FMT, a, b, c) \
+  __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4((a), (b), (c), (FMT), (FMT), 0, 127, 0, 127)
 
+template <int FMT>  // 0 = fp8 e4m3, 4 = fp4 e2m1 (nibble-packed)
 __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
-    const unsigned char* __restrict__ A,   // [M][K] e4m3
-    const unsigned char* __restrict__ Bt,  // [N][K] e4m3
+    const unsigned char* __restrict__ A,   // [M][K] packed
+    const unsigned char* __restrict__ Bt,  // [N][K] packed
     float* __restrict__ C,                 // [M][N]
-    int M, int N, int K) {
+    int M, int N, int K) {                 // K in ELEMENTS
   __shared__ unsigned char lds[SLOTS * HALF_BYTES];  // 128 KiB
 
   const int tiles_n = N / BN;
@@ -108,11 +126,13 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
   const int wn = wave & 3;
 
   const int frow = lane & 15;
-  const int fkb = (lane >> 4) * 32;  // 32-byte k-slice per lane
+  const int fkb = (lane >> 4) * 32;   // fp8: 32-byte k-slice per lane
+  const int fkb4 = (lane >> 4) * 16;  // fp4: 16-byte slice per k-half
 
-  const unsigned char* Ablk = A + (long)bm * K;
-  const unsigned char* Bblk = Bt + (long)bn * K;
-  const int T = K / BK;
+  const int Kb = (FMT == 4) ? K / 2 : K;   // row stride in bytes
+  const unsigned char* Ablk = A + (long)bm * Kb;
+  const unsigned char* Bblk = Bt + (long)bn * Kb;
+  const int T = Kb / BKB;
 
   f32x4 acc[2][4][2][2] = {};
 
@@ -121,14 +141,14 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
     int tt = h >> 2;
     const unsigned char* g;
     if (c == 0)
-      g = Ablk + tt * BK;
+      g = Ablk + tt * BKB;
     else if (c == 1)
-      g = Bblk + tt * BK;
+      g = Bblk + tt * BKB;
     else if (c == 2)
-      g = Ablk + (long)128 * K + tt * BK;
+      g = Ablk + (long)128 * Kb + tt * BKB;
     else
-      g = Bblk + (long)128 * K + tt * BK;
-    stage_half(g, K, lds + (h & 7) * HALF_BYTES);
+      g = Bblk + (long)128 * Kb + tt * BKB;
+    stage_half(g, Kb, lds + (h & 7) * HALF_BYTES);
   };
 
   for (int h = 0; h < 4 && h < 4 * T; ++h) stage_h(h);
@@ -139,6 +159,8 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
 
   i32x8 afrag[4];
   i32x8 bfrag[2][2];
+  i32x8 afrag4[4][2];     // fp4: one 16-B fragment per 128-element k-half
+  i32x8 bfrag4[2][2][2];
 
   const int arow = wm * 64 + frow;
   const int brow = wn * 32 + frow;
@@ -150,12 +172,25 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
     const unsigned char* Bs1 = lds + ((4 * t + 3) & 7) * HALF_BYTES;
 
     // ---- phase 0: q(0,0); read A(qm0)+B set 0; stage 4t+7
+    if (FMT == 4) {
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
-      afrag[fm] = read_frag(As0, arow + fm * 16, fkb);
+      for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-    for (int fn = 0; fn < 2; ++fn)
-      bfrag[0][fn] = read_frag(Bs0, brow + fn * 16, fkb);
+        for (int ks = 0; ks < 2; ++ks)
+          afrag4[fm][ks] = read_frag4(As0, arow + fm * 16, ks * 64 + fkb4);
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          bfrag4[0][fn][ks] = read_frag4(Bs0, brow + fn * 16, ks * 64 + fkb4);
+    } else {
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+        afrag[fm] = read_frag(As0, arow + fm * 16, fkb);
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+        bfrag[0][fn] = read_frag(Bs0, brow + fn * 16, fkb);
+    }
     if (4 * t + 7 < 4 * T) stage_h(4 * t + 7);
     asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
     __builtin_amdgcn_s_barrier();
@@ -165,14 +200,29 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
-        acc[0][fm][0][fn] = MX_MFMA(afrag[fm], bfrag[0][fn], acc[0][fm][0][fn]);
+        if (FMT == 4) {
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[0][fm][0][fn] = MX_MFMA(4, afrag4[fm][ks], bfrag4[0][fn][ks],
+                                          acc[0][fm][0][fn]);
+        } else {
+          acc[0][fm][0][fn] = MX_MFMA(0, afrag[fm], bfrag[0][fn], acc[0][fm][0][fn]);
+        }
     __builtin_amdgcn_s_setprio(0);
     __builtin_amdgcn_s_barrier();
 
     // ---- phase 1: q(0,1); read B set 1; stage 4t+8
+    if (FMT == 4) {
 #pragma unroll
-    for (int fn = 0; fn < 2; ++fn)
-      bfrag[1][fn] = read_frag(Bs1, brow + fn * 16, fkb);
+      for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          bfrag4[1][fn][ks] = read_frag4(Bs1, brow + fn * 16, ks * 64 + fkb4);
+    } else {
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+        bfrag[1][fn] = read_frag(Bs1, brow + fn * 16, fkb);
+    }
     if (4 * t + 8 < 4 * T) stage_h(4 * t + 8);
     __builtin_amdgcn_s_barrier();
     wait_lgkm0_fence();
@@ -181,14 +231,29 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
-        acc[0][fm][1][fn] = MX_MFMA(afrag[fm], bfrag[1][fn], acc[0][fm][1][fn]);
+        if (FMT == 4) {
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[0][fm][1][fn] = MX_MFMA(4, afrag4[fm][ks], bfrag4[1][fn][ks],
+                                          acc[0][fm][1][fn]);
+        } else {
+          acc[0][fm][1][fn] = MX_MFMA(0, afrag[fm], bfrag[1][fn], acc[0][fm][1][fn]);
+        }
     __builtin_amdgcn_s_setprio(0);
     __builtin_amdgcn_s_barrier();
 
     // ---- phase 2: q(1,1); re-read A(qm1); stage 4t+9
+    if (FMT == 4) {
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
-      afrag[fm] = read_frag(As1, arow + fm * 16, fkb);
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          afrag4[fm][ks] = read_frag4(As1, arow + fm * 16, ks * 64 + fkb4);
+    } else {
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+        afrag[fm] = read_frag(As1, arow + fm * 16, fkb);
+    }
     if (4 * t + 9 < 4 * T) stage_h(4 * t + 9);
     __builtin_amdgcn_s_barrier();
     wait_lgkm0_fence();
@@ -197,7 +262,14 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
-        acc[1][fm][1][fn] = MX_MFMA(afrag[fm], bfrag[1][fn], acc[1][fm][1][fn]);
+        if (FMT == 4) {
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[1][fm][1][fn] = MX_MFMA(4, afrag4[fm][ks], bfrag4[1][fn][ks],
+                                          acc[1][fm][1][fn]);
+        } else {
+          acc[1][fm][1][fn] = MX_MFMA(0, afrag[fm], bfrag[1][fn], acc[1][fm][1][fn]);
+        }
     __builtin_amdgcn_s_setprio(0);
     __builtin_amdgcn_s_barrier();
 
@@ -210,7 +282,14 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
-        acc[1][fm][0][fn] = MX_MFMA(afrag[fm], bfrag[0][fn], acc[1][fm][0][fn]);
+        if (FMT == 4) {
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[1][fm][0][fn] = MX_MFMA(4, afrag4[fm][ks], bfrag4[0][fn][ks],
+                                          acc[1][fm][0][fn]);
+        } else {
+          acc[1][fm][0][fn] = MX_MFMA(0, afrag[fm], bfrag[0][fn], acc[1][fm][0][fn]);
+        }
     __builtin_amdgcn_s_setprio(0);
     {
       int staged = min(4 * T, 4 * t + 11);
@@ -253,6 +332,19 @@ __global__ void fill_e4m3_hash_kernel(unsigned char* p, size_t n, unsigned seed)
     unsigned char b = (unsigned char)(((h & 1u) << 7) | ((((h >> 1) & 7u) + 4u) << 3) |
                                       ((h >> 4) & 7u));
     p[i] = b;
+  }
+}
+
+// pseudorandom e2m1 (fp4) nibble fill: all 16 codes are valid values
+__global__ void fill_e2m1_hash_kernel(unsigned char* p, size_t n, unsigned seed) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    unsigned h = (unsigned)(i * 2654435761u) ^ seed;
+    h ^= h >> 13;
+    h *= 0x85ebca6bu;
+    h ^= h >> 16;
+    p[i] = (unsigned char)(h & 0xFF);
   }
 }
 

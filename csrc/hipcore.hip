@@ -681,7 +681,30 @@ torch::Tensor gemm_fp8_mx_bt(torch::Tensor A, torch::Tensor Bt) {
   auto C = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
   int grid = (M / 256) * (N / 256);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel, dim3(grid), dim3(512),
+  hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<0>, dim3(grid), dim3(512),
+                     0, stream.stream(),
+                     reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                     reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                     C.data_ptr<float>(), M, N, K);
+  return C;
+}
+
+torch::Tensor gemm_fp4_mx_bt(torch::Tensor A, torch::Tensor Bt, int K) {
+  // A [M][K/2], Bt [N][K/2]: e2m1 nibble-packed (low nibble = even k)
+  TORCH_CHECK(A.is_cuda() && Bt.is_cuda(), "GPU tensors required");
+  TORCH_CHECK(A.scalar_type() == torch::kUInt8 &&
+              Bt.scalar_type() == torch::kUInt8, "packed e2m1 bytes expected");
+  A = A.contiguous();
+  Bt = Bt.contiguous();
+  int M = A.size(0), N = Bt.size(0);
+  TORCH_CHECK((long)A.size(1) * 2 == K && (long)Bt.size(1) * 2 == K,
+              "K must equal 2 x packed byte columns");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 256 == 0 && K >= 512,
+              "M,N multiples of 256; K multiple of 256, >= 512");
+  auto C = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
+  int grid = (M / 256) * (N / 256);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<4>, dim3(grid), dim3(512),
                      0, stream.stream(),
                      reinterpret_cast<const unsigned char*>(A.data_ptr()),
                      reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
@@ -707,7 +730,35 @@ double gemm_fp8_mx_tflops(int device, int size, int iters) {
   double ms = time_kernel_ms(
       device,
       [&](hipStream_t s) {
-        hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel, dim3(grid),
+        hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<0>, dim3(grid),
+                           dim3(512), 0, s, A, Bt, C, size, size, size);
+      },
+      iters);
+  (void)hipFree(A);
+  (void)hipFree(Bt);
+  (void)hipFree(C);
+  return 2.0 * size * (double)size * size / (ms * 1e9);
+}
+
+double gemm_fp4_mx_tflops(int device, int size, int iters) {
+  HIP_CHECK(hipSetDevice(device));
+  mfma_warmup(device, 20000);
+  size_t nb = (size_t)size * size / 2;  // nibble-packed
+  unsigned char *A = nullptr, *Bt = nullptr;
+  float* C = nullptr;
+  HIP_CHECK(hipMalloc(&A, nb));
+  HIP_CHECK(hipMalloc(&Bt, nb));
+  HIP_CHECK(hipMalloc(&C, (size_t)size * size * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fp8_mx::fill_e2m1_hash_kernel, dim3(4096), dim3(256),
+                     0, 0, A, nb, 1u);
+  hipLaunchKernelGGL(gemm_fp8_mx::fill_e2m1_hash_kernel, dim3(4096), dim3(256),
+                     0, 0, Bt, nb, 7u);
+  HIP_CHECK(hipDeviceSynchronize());
+  int grid = (size / 256) * (size / 256);
+  double ms = time_kernel_ms(
+      device,
+      [&](hipStream_t s) {
+        hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<4>, dim3(grid),
                            dim3(512), 0, s, A, Bt, C, size, size, size);
       },
       iters);
@@ -756,6 +807,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16_8ph", &gemm_bf16_8ph_bt, py::arg("A"), py::arg("Bt"),
         py::arg("variant") = 0);
   m.def("gemm_fp8_mx", &gemm_fp8_mx_bt, py::arg("A"), py::arg("Bt"));
+  m.def("gemm_fp4_mx", &gemm_fp4_mx_bt, py::arg("A"), py::arg("Bt"), py::arg("K"));
+  m.def("gemm_fp4_mx_tflops", &gemm_fp4_mx_tflops, py::arg("device") = 0,
+        py::arg("size") = 4096, py::arg("iters") = 10,
+        py::call_guard<py::gil_scoped_release>());
   m.def("gemm_fp8_mx_tflops", &gemm_fp8_mx_tflops, py::arg("device") = 0,
         py::arg("size") = 4096, py::arg("iters") = 10,
         py::call_guard<py::gil_scoped_release>());

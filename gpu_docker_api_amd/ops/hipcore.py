@@ -119,6 +119,8 @@ def validate_gpus(
         }
         if fp8 and size % 256 == 0 and size >= 256:
             entry["fp8_tflops"] = round(ext.gemm_fp8_mx_tflops(i, size, iters), 1)
+        if fp8 and size % 256 == 0 and size >= 512:
+            entry["fp4_tflops"] = round(ext.gemm_fp4_mx_tflops(i, size, iters), 1)
         report["gpus"].append(entry)
     vals = [g["bf16_tflops"] for g in report["gpus"]]
     if vals:

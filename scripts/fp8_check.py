@@ -30,6 +30,29 @@ def main():
         out[f"tflops_{size}"] = [
             round(ext.gemm_fp8_mx_tflops(0, size, iters), 1) for _ in range(3)
         ]
+
+    # fp4 (e2m1): pack random nibbles, dequant via LUT for the reference
+    lut = torch.tensor(
+        [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
+         -0.0, -0.5, -1.0, -1.5, -2.0, -3.0, -4.0, -6.0], device="cuda")
+    for (M, N, K) in ((256, 256, 512), (512, 256, 1024), (4096, 4096, 4096)):
+        gen = torch.Generator(device="cuda").manual_seed(13)
+        nibbles = torch.randint(0, 16, (M, K), generator=gen, device="cuda", dtype=torch.uint8)
+        nibB = torch.randint(0, 16, (N, K), generator=gen, device="cuda", dtype=torch.uint8)
+        packA = (nibbles[:, 0::2] | (nibbles[:, 1::2] << 4)).contiguous()
+        packB = (nibB[:, 0::2] | (nibB[:, 1::2] << 4)).contiguous()
+        ref = lut[nibbles.long()] @ lut[nibB.long()].T
+        scale = ref.abs().max().item() + 1e-6
+        worst = 0.0
+        for _ in range(4):
+            C = ext.gemm_fp4_mx(packA, packB, K)
+            torch.cuda.synchronize()
+            worst = max(worst, (C - ref).abs().max().item() / scale)
+        out[f"fp4_relerr_{M}x{N}x{K}"] = worst
+    for size, iters in ((4096, 8), (8192, 4)):
+        out[f"fp4_tflops_{size}"] = [
+            round(ext.gemm_fp4_mx_tflops(0, size, iters), 1) for _ in range(3)
+        ]
     print(json.dumps(out))
     os.makedirs("gpurun_out", exist_ok=True)
     open("gpurun_out/fp8_mx.json", "w").write(json.dumps(out))

@@ -329,3 +329,28 @@ def test_gemm_fp8_mx_numerics_and_throughput(ext):
     tflops = ext.gemm_fp8_mx_tflops(0, 4096, 8)
     print(f"fp8 MX GEMM (8-phase 256^2, K=128): {tflops:.0f} TFLOPS @4096^3")
     assert tflops > 1200, f"fp8 MX GEMM regressed: {tflops} TF (floor 1200 = 75% of measured 1666)"
+
+
+def test_gemm_fp4_mx_numerics_and_throughput(ext):
+    """MX-fp4 (e2m1, nibble-packed, K-step 256): every e2m1 value is a
+    small multiple of 0.5, so fp32 accumulation is EXACT — the kernel must
+    match the LUT-dequantized torch reference bit-for-bit. Measured round
+    2: 3086-3213 TF @4096^3, ~4.0 PF @8192^3 (profiles/gemm_fp8_mx.json)."""
+    import torch
+
+    lut = torch.tensor(
+        [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
+         -0.0, -0.5, -1.0, -1.5, -2.0, -3.0, -4.0, -6.0], device="cuda")
+    gen = torch.Generator(device="cuda").manual_seed(13)
+    nA = torch.randint(0, 16, (512, 1024), generator=gen, device="cuda", dtype=torch.uint8)
+    nB = torch.randint(0, 16, (256, 1024), generator=gen, device="cuda", dtype=torch.uint8)
+    packA = (nA[:, 0::2] | (nA[:, 1::2] << 4)).contiguous()
+    packB = (nB[:, 0::2] | (nB[:, 1::2] << 4)).contiguous()
+    ref = lut[nA.long()] @ lut[nB.long()].T
+    for _ in range(3):  # race screen
+        C = ext.gemm_fp4_mx(packA, packB, 1024)
+        torch.cuda.synchronize()
+        assert torch.equal(C, ref), "fp4 MX result not exact"
+    tflops = ext.gemm_fp4_mx_tflops(0, 4096, 8)
+    print(f"fp4 MX GEMM: {tflops:.0f} TFLOPS @4096^3")
+    assert tflops > 2300, f"fp4 MX GEMM regressed: {tflops} TF (floor 2300 = 75% of measured 3086)"
