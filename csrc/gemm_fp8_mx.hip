@@ -38,6 +38,7 @@
 namespace gemm_fp8_mx {
 
 using f32x4 = __attribute__((ext_vector_type(4))) float;
+using f32x16 = __attribute__((ext_vector_type(16))) float;
 using i32x4 = __attribute__((ext_vector_type(4))) int;
 using i32x8 = __attribute__((ext_vector_type(8))) int;
 
@@ -315,6 +316,179 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
           for (int r = 0; r < 4; ++r)
             C[(long)(row0 + r) * N + col] = acc[qm][fm][qn][fn][r];
         }
+}
+
+
+#define MX_MFMA32(FMT, a, b, c) \
+  __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4((a), (b), (c), (FMT), (FMT), 0, 127, 0, 127)
+
+// 32x32x64 variant: same 256^2 tile / 8-wave / 8-phase schedule and the
+// same 16 KiB half-tile staging; fragments are 32-row/32-col, each phase
+// runs {fp8: 2 m-frags x 2 k-steps, fp4: 2 x 4} MFMAs of the 32x32x64
+// scaled instruction (~25% higher uarch ceiling than 16x16x128 — guide §3
+// µbench 9099 vs 7228 TF fp4). Operand/C-D layouts hardware-verified by
+// csrc-probe (max abs err 0.0): A lane l = row l&31, k (l>>5)*32 elements;
+// C col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5).
+template <int FMT>  // 0 = fp8 e4m3, 4 = fp4 e2m1
+__global__ __launch_bounds__(THREADS, 2) void gemm_mx32_kernel(
+    const unsigned char* __restrict__ A,   // [M][K] packed
+    const unsigned char* __restrict__ Bt,  // [N][K] packed
+    float* __restrict__ C,                 // [M][N]
+    int M, int N, int K) {                 // K in ELEMENTS
+  __shared__ unsigned char lds[SLOTS * HALF_BYTES];  // 128 KiB
+
+  constexpr int KSTEPS = (FMT == 4) ? 4 : 2;     // 32B(fp4)/64B(fp8) per step
+  constexpr int STEPB = BKB / KSTEPS;            // bytes per k-step row-slice
+  constexpr int LANEB = STEPB / 2;               // per-lane slice (l>>5 half)
+
+  const int tiles_n = N / BN;
+  const int bm = ((int)blockIdx.x / tiles_n) * BM;
+  const int bn = ((int)blockIdx.x % tiles_n) * BN;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+
+  const int frow = lane & 31;
+  const int fkb = (lane >> 5) * LANEB;
+
+  const int Kb = (FMT == 4) ? K / 2 : K;
+  const unsigned char* Ablk = A + (long)bm * Kb;
+  const unsigned char* Bblk = Bt + (long)bn * Kb;
+  const int T = Kb / BKB;
+
+  f32x16 acc[2][2][2] = {};  // [qm][fm][qn]; fn==0 (one 32-col frag/wave)
+
+  auto stage_h = [&](int h) {
+    int c = h & 3;
+    int tt = h >> 2;
+    const unsigned char* g;
+    if (c == 0)
+      g = Ablk + tt * BKB;
+    else if (c == 1)
+      g = Bblk + tt * BKB;
+    else if (c == 2)
+      g = Ablk + (long)128 * Kb + tt * BKB;
+    else
+      g = Bblk + (long)128 * Kb + tt * BKB;
+    stage_half(g, Kb, lds + (h & 7) * HALF_BYTES);
+  };
+
+  for (int h = 0; h < 4 && h < 4 * T; ++h) stage_h(h);
+  wait_vmcnt(2);
+  for (int h = 4; h < 7 && h < 4 * T; ++h) stage_h(h);
+  wait_vmcnt(3);
+  __builtin_amdgcn_s_barrier();
+
+  i32x8 afrag[2][KSTEPS];
+  i32x8 bfrag[2][KSTEPS];  // [set][ks]
+
+  const int arow = wm * 64 + frow;
+  const int brow = wn * 32 + frow;
+
+  auto read_f = [&](const unsigned char* base, int row, int ks) -> i32x8 {
+    if (FMT == 4) return read_frag4(base, row, ks * STEPB + fkb);
+    return read_frag(base, row, ks * STEPB + fkb);
+  };
+
+  for (int t = 0; t < T; ++t) {
+    const unsigned char* As0 = lds + ((4 * t + 0) & 7) * HALF_BYTES;
+    const unsigned char* Bs0 = lds + ((4 * t + 1) & 7) * HALF_BYTES;
+    const unsigned char* As1 = lds + ((4 * t + 2) & 7) * HALF_BYTES;
+    const unsigned char* Bs1 = lds + ((4 * t + 3) & 7) * HALF_BYTES;
+
+    // ---- phase 0: q(0,0)
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks)
+        afrag[fm][ks] = read_f(As0, arow + fm * 32, ks);
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks)
+      bfrag[0][ks] = read_f(Bs0, brow, ks);
+    if (4 * t + 7 < 4 * T) stage_h(4 * t + 7);
+    asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    wait_lgkm0_fence();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks)
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+        acc[0][fm][0] = MX_MFMA32(FMT, afrag[fm][ks], bfrag[0][ks], acc[0][fm][0]);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+
+    // ---- phase 1: q(0,1)
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks)
+      bfrag[1][ks] = read_f(Bs1, brow, ks);
+    if (4 * t + 8 < 4 * T) stage_h(4 * t + 8);
+    __builtin_amdgcn_s_barrier();
+    wait_lgkm0_fence();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks)
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+        acc[0][fm][1] = MX_MFMA32(FMT, afrag[fm][ks], bfrag[1][ks], acc[0][fm][1]);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+
+    // ---- phase 2: q(1,1)
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks)
+        afrag[fm][ks] = read_f(As1, arow + fm * 32, ks);
+    if (4 * t + 9 < 4 * T) stage_h(4 * t + 9);
+    __builtin_amdgcn_s_barrier();
+    wait_lgkm0_fence();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks)
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+        acc[1][fm][1] = MX_MFMA32(FMT, afrag[fm][ks], bfrag[1][ks], acc[1][fm][1]);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+
+    // ---- phase 3: q(1,0)
+    if (4 * t + 10 < 4 * T) stage_h(4 * t + 10);
+    __builtin_amdgcn_s_barrier();
+    wait_lgkm0_fence();
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks)
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+        acc[1][fm][0] = MX_MFMA32(FMT, afrag[fm][ks], bfrag[0][ks], acc[1][fm][0]);
+    __builtin_amdgcn_s_setprio(0);
+    {
+      int staged = min(4 * T, 4 * t + 11);
+      int allowed = staged - 4 * (t + 2);
+      wait_vmcnt(allowed < 0 ? 0 : (allowed > 3 ? 3 : allowed));
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue: 32x32 C/D map
+  const int ccol = lane & 31;
+  const int rbase = 4 * (lane >> 5);
+#pragma unroll
+  for (int qm = 0; qm < 2; ++qm)
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+      for (int qn = 0; qn < 2; ++qn) {
+        int row0 = bm + qm * 128 + wm * 64 + fm * 32;
+        int col = bn + qn * 128 + wn * 32 + ccol;
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+          int r = (reg & 3) + 8 * (reg >> 2) + rbase;
+          C[(long)(row0 + r) * N + col] = acc[qm][fm][qn][reg];
+        }
+      }
 }
 
 // pseudorandom VALID e4m3 fill: full sign/mantissa variation, exponents

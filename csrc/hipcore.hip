@@ -667,7 +667,7 @@ py::dict gemm_bf16_8ph_ab(int device, int size, int iters, int rounds,
   return d;
 }
 
-torch::Tensor gemm_fp8_mx_bt(torch::Tensor A, torch::Tensor Bt) {
+torch::Tensor gemm_fp8_mx_bt(torch::Tensor A, torch::Tensor Bt, int shape) {
   TORCH_CHECK(A.is_cuda() && Bt.is_cuda(), "GPU tensors required");
   TORCH_CHECK(A.scalar_type() == torch::kUInt8 &&
               Bt.scalar_type() == torch::kUInt8,
@@ -681,15 +681,22 @@ torch::Tensor gemm_fp8_mx_bt(torch::Tensor A, torch::Tensor Bt) {
   auto C = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
   int grid = (M / 256) * (N / 256);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<0>, dim3(grid), dim3(512),
-                     0, stream.stream(),
-                     reinterpret_cast<const unsigned char*>(A.data_ptr()),
-                     reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
-                     C.data_ptr<float>(), M, N, K);
+  if (shape == 32)
+    hipLaunchKernelGGL(gemm_fp8_mx::gemm_mx32_kernel<0>, dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
+  else
+    hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<0>, dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
   return C;
 }
 
-torch::Tensor gemm_fp4_mx_bt(torch::Tensor A, torch::Tensor Bt, int K) {
+torch::Tensor gemm_fp4_mx_bt(torch::Tensor A, torch::Tensor Bt, int K, int shape) {
   // A [M][K/2], Bt [N][K/2]: e2m1 nibble-packed (low nibble = even k)
   TORCH_CHECK(A.is_cuda() && Bt.is_cuda(), "GPU tensors required");
   TORCH_CHECK(A.scalar_type() == torch::kUInt8 &&
@@ -704,15 +711,22 @@ torch::Tensor gemm_fp4_mx_bt(torch::Tensor A, torch::Tensor Bt, int K) {
   auto C = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
   int grid = (M / 256) * (N / 256);
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<4>, dim3(grid), dim3(512),
-                     0, stream.stream(),
-                     reinterpret_cast<const unsigned char*>(A.data_ptr()),
-                     reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
-                     C.data_ptr<float>(), M, N, K);
+  if (shape == 32)
+    hipLaunchKernelGGL(gemm_fp8_mx::gemm_mx32_kernel<4>, dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
+  else
+    hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<4>, dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
   return C;
 }
 
-double gemm_fp8_mx_tflops(int device, int size, int iters) {
+double gemm_fp8_mx_tflops(int device, int size, int iters, int shape) {
   HIP_CHECK(hipSetDevice(device));
   mfma_warmup(device, 20000);
   size_t n = (size_t)size * size;
@@ -730,8 +744,12 @@ double gemm_fp8_mx_tflops(int device, int size, int iters) {
   double ms = time_kernel_ms(
       device,
       [&](hipStream_t s) {
-        hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<0>, dim3(grid),
-                           dim3(512), 0, s, A, Bt, C, size, size, size);
+        if (shape == 32)
+          hipLaunchKernelGGL(gemm_fp8_mx::gemm_mx32_kernel<0>, dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
+        else
+          hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<0>, dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
       },
       iters);
   (void)hipFree(A);
@@ -740,7 +758,7 @@ double gemm_fp8_mx_tflops(int device, int size, int iters) {
   return 2.0 * size * (double)size * size / (ms * 1e9);
 }
 
-double gemm_fp4_mx_tflops(int device, int size, int iters) {
+double gemm_fp4_mx_tflops(int device, int size, int iters, int shape) {
   HIP_CHECK(hipSetDevice(device));
   mfma_warmup(device, 20000);
   size_t nb = (size_t)size * size / 2;  // nibble-packed
@@ -758,8 +776,12 @@ double gemm_fp4_mx_tflops(int device, int size, int iters) {
   double ms = time_kernel_ms(
       device,
       [&](hipStream_t s) {
-        hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<4>, dim3(grid),
-                           dim3(512), 0, s, A, Bt, C, size, size, size);
+        if (shape == 32)
+          hipLaunchKernelGGL(gemm_fp8_mx::gemm_mx32_kernel<4>, dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
+        else
+          hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<4>, dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
       },
       iters);
   (void)hipFree(A);
@@ -806,13 +828,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16_bt", &gemm_bf16_bt, py::arg("A"), py::arg("Bt"));
   m.def("gemm_bf16_8ph", &gemm_bf16_8ph_bt, py::arg("A"), py::arg("Bt"),
         py::arg("variant") = 0);
-  m.def("gemm_fp8_mx", &gemm_fp8_mx_bt, py::arg("A"), py::arg("Bt"));
-  m.def("gemm_fp4_mx", &gemm_fp4_mx_bt, py::arg("A"), py::arg("Bt"), py::arg("K"));
+  m.def("gemm_fp8_mx", &gemm_fp8_mx_bt, py::arg("A"), py::arg("Bt"),
+        py::arg("shape") = 16);
+  m.def("gemm_fp4_mx", &gemm_fp4_mx_bt, py::arg("A"), py::arg("Bt"), py::arg("K"),
+        py::arg("shape") = 16);
   m.def("gemm_fp4_mx_tflops", &gemm_fp4_mx_tflops, py::arg("device") = 0,
-        py::arg("size") = 4096, py::arg("iters") = 10,
+        py::arg("size") = 4096, py::arg("iters") = 10, py::arg("shape") = 16,
         py::call_guard<py::gil_scoped_release>());
   m.def("gemm_fp8_mx_tflops", &gemm_fp8_mx_tflops, py::arg("device") = 0,
-        py::arg("size") = 4096, py::arg("iters") = 10,
+        py::arg("size") = 4096, py::arg("iters") = 10, py::arg("shape") = 16,
         py::call_guard<py::gil_scoped_release>());
   m.def("gemm_bf16_8ph_ab", &gemm_bf16_8ph_ab, py::arg("device") = 0,
         py::arg("size") = 4096, py::arg("iters") = 4, py::arg("rounds") = 3,

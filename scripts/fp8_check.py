@@ -20,16 +20,21 @@ def main():
         Bt = (torch.randn(N, K, device="cuda") * 0.5).to(torch.float8_e4m3fn)
         ref = A.float() @ Bt.float().T
         scale = ref.abs().max().item() + 1e-6
-        worst = 0.0
-        for _ in range(4):
-            C = ext.gemm_fp8_mx(A.view(torch.uint8), Bt.view(torch.uint8))
-            torch.cuda.synchronize()
-            worst = max(worst, (C - ref).abs().max().item() / scale)
-        out[f"relerr_{M}x{N}x{K}"] = worst
+        for shape in (16, 32):
+            worst = 0.0
+            for _ in range(4):
+                C = ext.gemm_fp8_mx(A.view(torch.uint8), Bt.view(torch.uint8), shape=shape)
+                torch.cuda.synchronize()
+                worst = max(worst, (C - ref).abs().max().item() / scale)
+            out[f"relerr_s{shape}_{M}x{N}x{K}"] = worst
     for size, iters in ((4096, 8), (8192, 4)):
-        out[f"tflops_{size}"] = [
-            round(ext.gemm_fp8_mx_tflops(0, size, iters), 1) for _ in range(3)
-        ]
+        # interleaved 16 vs 32 shape A/B
+        for shape in (16, 32):
+            out[f"tflops_s{shape}_{size}"] = []
+        for _ in range(3):
+            for shape in (16, 32):
+                out[f"tflops_s{shape}_{size}"].append(
+                    round(ext.gemm_fp8_mx_tflops(0, size, iters, shape=shape), 1))
 
     # fp4 (e2m1): pack random nibbles, dequant via LUT for the reference
     lut = torch.tensor(
@@ -43,16 +48,20 @@ def main():
         packB = (nibB[:, 0::2] | (nibB[:, 1::2] << 4)).contiguous()
         ref = lut[nibbles.long()] @ lut[nibB.long()].T
         scale = ref.abs().max().item() + 1e-6
-        worst = 0.0
-        for _ in range(4):
-            C = ext.gemm_fp4_mx(packA, packB, K)
-            torch.cuda.synchronize()
-            worst = max(worst, (C - ref).abs().max().item() / scale)
-        out[f"fp4_relerr_{M}x{N}x{K}"] = worst
+        for shape in (16, 32):
+            worst = 0.0
+            for _ in range(4):
+                C = ext.gemm_fp4_mx(packA, packB, K, shape=shape)
+                torch.cuda.synchronize()
+                worst = max(worst, (C - ref).abs().max().item() / scale)
+            out[f"fp4_relerr_s{shape}_{M}x{N}x{K}"] = worst
     for size, iters in ((4096, 8), (8192, 4)):
-        out[f"fp4_tflops_{size}"] = [
-            round(ext.gemm_fp4_mx_tflops(0, size, iters), 1) for _ in range(3)
-        ]
+        for shape in (16, 32):
+            out[f"fp4_tflops_s{shape}_{size}"] = []
+        for _ in range(3):
+            for shape in (16, 32):
+                out[f"fp4_tflops_s{shape}_{size}"].append(
+                    round(ext.gemm_fp4_mx_tflops(0, size, iters, shape=shape), 1))
     print(json.dumps(out))
     os.makedirs("gpurun_out", exist_ok=True)
     open("gpurun_out/fp8_mx.json", "w").write(json.dumps(out))
