@@ -6,7 +6,7 @@ PY ?= python3
 HIPCC ?= /opt/rocm/bin/hipcc
 ARCH ?= gfx950
 
-.PHONY: all native test test-gpu bench run run-mock clean openapi
+.PHONY: all native test test-gpu bench run run-mock clean openapi parity run-docker-sim run-etcd-sim
 
 all: native
 
@@ -33,6 +33,15 @@ run-mock:          ## daemon in full-mock mode (CPU-only dev box)
 
 openapi:           ## regenerate api/openapi.json from the live app
 	$(PY) scripts/export_openapi.py
+
+parity:            ## prove route/body parity vs the reference spec
+	$(PY) scripts/openapi_parity.py
+
+run-docker-sim:    ## dev engine: dockerd-compatible API over a unix socket
+	$(PY) -m gpu_docker_api_amd.testing.dockerd_sim --socket /tmp/gda-dockerd.sock --data /tmp/gda-engine
+
+run-etcd-sim:      ## dev etcd: v3 JSON gateway on TCP
+	$(PY) -m gpu_docker_api_amd.state.etcd_fake --port 2379
 
 clean:
 	rm -rf gpu_docker_api_amd/ops/*.so csrc/bin .state merges
