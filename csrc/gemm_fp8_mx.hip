@@ -55,12 +55,25 @@ __device__ inline void glds16(const unsigned char* gsrc, unsigned char* lds) {
       (__attribute__((address_space(3))) unsigned int*)lds, 16, 0, 0);
 }
 
-// chunk-xor swizzle in 16-B granules (byte ^= (row&7)<<4) — the form that
-// measured conflict-free on the bf16 sibling (same 128-B row stride)
-__device__ inline int swz_byte(int row, int byte_off) {
-  return byte_off ^ ((row & 7) << 4);
+// chunk-xor swizzle in 16-B granules — SWZV selects the row hash g(row):
+//   0: g = row&7 (the bf16 sibling's form; fp8's two-chunk fragment reads
+//      measured 44M SQ_LDS_BANK_CONFLICT/6-dispatch with it — rows r and
+//      r+8 always collide under a 3-bit hash of row&7)
+//   1: g = (row&7) ^ ((row>>3)&1) — folds row bit 3 into the hash, making
+//      the 16 rows of a b128 lane group land on 16 disjoint 4-bank spans
+//      (derivation in profiles/gemm_kernel_stats.md). Any per-row xor is
+//      self-inverse, so staging source and reads share the formula.
+template <int SWZV>
+__device__ inline int swz_byte_v(int row, int byte_off) {
+  int g = (row & 7) ^ (SWZV ? ((row >> 3) & 1) : 0);
+  return byte_off ^ (g << 4);
 }
 
+__device__ inline int swz_byte(int row, int byte_off) {
+  return swz_byte_v<0>(row, byte_off);
+}
+
+template <int SWZV>
 __device__ inline void stage_half(const unsigned char* gbase, int ldk,
                                   unsigned char* half_base) {
   const int t = threadIdx.x;
@@ -69,26 +82,28 @@ __device__ inline void stage_half(const unsigned char* gbase, int ldk,
     int slot = pass * THREADS + t;
     int row_l = slot >> 3;
     int byte_off = (slot & 7) * 16;
-    glds16(gbase + row_l * ldk + swz_byte(row_l, byte_off),
+    glds16(gbase + row_l * ldk + swz_byte_v<SWZV>(row_l, byte_off),
            half_base + row_l * BKB + byte_off);
   }
 }
 
 // one fp8 MFMA fragment (32 B): two independently-swizzled 16-B chunks
+template <int SWZV>
 __device__ inline i32x8 read_frag(const unsigned char* half_base, int row,
                                   int kb) {
   i32x4 lo = *reinterpret_cast<const i32x4*>(half_base + row * BKB +
-                                             swz_byte(row, kb));
+                                             swz_byte_v<SWZV>(row, kb));
   i32x4 hi = *reinterpret_cast<const i32x4*>(half_base + row * BKB +
-                                             swz_byte(row, kb + 16));
+                                             swz_byte_v<SWZV>(row, kb + 16));
   return i32x8{lo[0], lo[1], lo[2], lo[3], hi[0], hi[1], hi[2], hi[3]};
 }
 
 // one fp4 MFMA fragment (16 B = 32 nibbles) in the low 4 dwords
+template <int SWZV>
 __device__ inline i32x8 read_frag4(const unsigned char* half_base, int row,
                                    int kb) {
   i32x4 lo = *reinterpret_cast<const i32x4*>(half_base + row * BKB +
-                                             swz_byte(row, kb));
+                                             swz_byte_v<SWZV>(row, kb));
   return i32x8{lo[0], lo[1], lo[2], lo[3], 0, 0, 0, 0};
 }
 
@@ -110,7 +125,7 @@ __device__ inline void wait_vmcnt(int halves_outstanding) {
 #define MX_MFMA(FMT, a, b, c) \
   __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4((a), (b), (c), (FMT), (FMT), 0, 127, 0, 127)
 
-template <int FMT>  // 0 = fp8 e4m3, 4 = fp4 e2m1 (nibble-packed)
+template <int FMT, int SWZV = 0>  // FMT: 0 = fp8 e4m3, 4 = fp4 e2m1 (packed)
 __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
     const unsigned char* __restrict__ A,   // [M][K] packed
     const unsigned char* __restrict__ Bt,  // [N][K] packed
@@ -149,7 +164,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
       g = Ablk + (long)128 * Kb + tt * BKB;
     else
       g = Bblk + (long)128 * Kb + tt * BKB;
-    stage_half(g, Kb, lds + (h & 7) * HALF_BYTES);
+    stage_half<SWZV>(g, Kb, lds + (h & 7) * HALF_BYTES);
   };
 
   for (int h = 0; h < 4 && h < 4 * T; ++h) stage_h(h);
@@ -178,19 +193,19 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
       for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
-          afrag4[fm][ks] = read_frag4(As0, arow + fm * 16, ks * 64 + fkb4);
+          afrag4[fm][ks] = read_frag4<SWZV>(As0, arow + fm * 16, ks * 64 + fkb4);
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
-          bfrag4[0][fn][ks] = read_frag4(Bs0, brow + fn * 16, ks * 64 + fkb4);
+          bfrag4[0][fn][ks] = read_frag4<SWZV>(Bs0, brow + fn * 16, ks * 64 + fkb4);
     } else {
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm)
-        afrag[fm] = read_frag(As0, arow + fm * 16, fkb);
+        afrag[fm] = read_frag<SWZV>(As0, arow + fm * 16, fkb);
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
-        bfrag[0][fn] = read_frag(Bs0, brow + fn * 16, fkb);
+        bfrag[0][fn] = read_frag<SWZV>(Bs0, brow + fn * 16, fkb);
     }
     if (4 * t + 7 < 4 * T) stage_h(4 * t + 7);
     asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
@@ -218,11 +233,11 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
       for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
-          bfrag4[1][fn][ks] = read_frag4(Bs1, brow + fn * 16, ks * 64 + fkb4);
+          bfrag4[1][fn][ks] = read_frag4<SWZV>(Bs1, brow + fn * 16, ks * 64 + fkb4);
     } else {
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
-        bfrag[1][fn] = read_frag(Bs1, brow + fn * 16, fkb);
+        bfrag[1][fn] = read_frag<SWZV>(Bs1, brow + fn * 16, fkb);
     }
     if (4 * t + 8 < 4 * T) stage_h(4 * t + 8);
     __builtin_amdgcn_s_barrier();
@@ -249,11 +264,11 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
       for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
-          afrag4[fm][ks] = read_frag4(As1, arow + fm * 16, ks * 64 + fkb4);
+          afrag4[fm][ks] = read_frag4<SWZV>(As1, arow + fm * 16, ks * 64 + fkb4);
     } else {
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm)
-        afrag[fm] = read_frag(As1, arow + fm * 16, fkb);
+        afrag[fm] = read_frag<SWZV>(As1, arow + fm * 16, fkb);
     }
     if (4 * t + 9 < 4 * T) stage_h(4 * t + 9);
     __builtin_amdgcn_s_barrier();
@@ -329,7 +344,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_fp8_mx_kernel(
 // µbench 9099 vs 7228 TF fp4). Operand/C-D layouts hardware-verified by
 // csrc-probe (max abs err 0.0): A lane l = row l&31, k (l>>5)*32 elements;
 // C col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5).
-template <int FMT>  // 0 = fp8 e4m3, 4 = fp4 e2m1
+template <int FMT, int SWZV = 0>  // FMT: 0 = fp8 e4m3, 4 = fp4 e2m1
 __global__ __launch_bounds__(THREADS, 2) void gemm_mx32_kernel(
     const unsigned char* __restrict__ A,   // [M][K] packed
     const unsigned char* __restrict__ Bt,  // [N][K] packed
@@ -371,7 +386,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_mx32_kernel(
       g = Ablk + (long)128 * Kb + tt * BKB;
     else
       g = Bblk + (long)128 * Kb + tt * BKB;
-    stage_half(g, Kb, lds + (h & 7) * HALF_BYTES);
+    stage_half<SWZV>(g, Kb, lds + (h & 7) * HALF_BYTES);
   };
 
   for (int h = 0; h < 4 && h < 4 * T; ++h) stage_h(h);
@@ -387,8 +402,8 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_mx32_kernel(
   const int brow = wn * 32 + frow;
 
   auto read_f = [&](const unsigned char* base, int row, int ks) -> i32x8 {
-    if (FMT == 4) return read_frag4(base, row, ks * STEPB + fkb);
-    return read_frag(base, row, ks * STEPB + fkb);
+    if (FMT == 4) return read_frag4<SWZV>(base, row, ks * STEPB + fkb);
+    return read_frag<SWZV>(base, row, ks * STEPB + fkb);
   };
 
   for (int t = 0; t < T; ++t) {

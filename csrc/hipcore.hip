@@ -682,13 +682,19 @@ torch::Tensor gemm_fp8_mx_bt(torch::Tensor A, torch::Tensor Bt, int shape) {
   int grid = (M / 256) * (N / 256);
   auto stream = at::hip::getCurrentHIPStream();
   if (shape == 32)
-    hipLaunchKernelGGL(gemm_fp8_mx::gemm_mx32_kernel<0>, dim3(grid), dim3(512),
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_mx32_kernel<0, 0>), dim3(grid), dim3(512),
                        0, stream.stream(),
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
                        C.data_ptr<float>(), M, N, K);
-  else
-    hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<0>, dim3(grid), dim3(512),
+  else if (shape == 17)  // legacy row&7-only swizzle (A/B reference)
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 0>), dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
+  else  // default: row-bit-3 swizzle (conflict-free for the 2-chunk reads, +11%)
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1>), dim3(grid), dim3(512),
                        0, stream.stream(),
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
@@ -712,13 +718,19 @@ torch::Tensor gemm_fp4_mx_bt(torch::Tensor A, torch::Tensor Bt, int K, int shape
   int grid = (M / 256) * (N / 256);
   auto stream = at::hip::getCurrentHIPStream();
   if (shape == 32)
-    hipLaunchKernelGGL(gemm_fp8_mx::gemm_mx32_kernel<4>, dim3(grid), dim3(512),
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_mx32_kernel<4, 0>), dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
+  else if (shape == 17)  // 16x16 with the row-bit-3 swizzle
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 1>), dim3(grid), dim3(512),
                        0, stream.stream(),
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
                        C.data_ptr<float>(), M, N, K);
   else
-    hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<4>, dim3(grid), dim3(512),
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0>), dim3(grid), dim3(512),
                        0, stream.stream(),
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
@@ -745,10 +757,13 @@ double gemm_fp8_mx_tflops(int device, int size, int iters, int shape) {
       device,
       [&](hipStream_t s) {
         if (shape == 32)
-          hipLaunchKernelGGL(gemm_fp8_mx::gemm_mx32_kernel<0>, dim3(grid),
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_mx32_kernel<0, 0>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
-        else
-          hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<0>, dim3(grid),
+        else if (shape == 17)  // legacy swizzle (A/B reference)
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 0>), dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
+        else  // default: row-bit-3 swizzle
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
       },
       iters);
@@ -777,10 +792,13 @@ double gemm_fp4_mx_tflops(int device, int size, int iters, int shape) {
       device,
       [&](hipStream_t s) {
         if (shape == 32)
-          hipLaunchKernelGGL(gemm_fp8_mx::gemm_mx32_kernel<4>, dim3(grid),
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_mx32_kernel<4, 0>), dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
+        else if (shape == 17)
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 1>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
         else
-          hipLaunchKernelGGL(gemm_fp8_mx::gemm_fp8_mx_kernel<4>, dim3(grid),
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
       },
       iters);
