@@ -93,11 +93,18 @@ __device__ inline void glds16(const __hip_bfloat16* gsrc, __hip_bfloat16* lds) {
 // the glds SOURCE (stage_half inverts it) and the ds_read address
 // (rule 21: both-sides-or-neither).
 __device__ inline int swz_row(int row) { return row; }
-__device__ inline int swz_k(int row, int k) { return k ^ ((row & 7) << 3); }
+template <int SWZV = 0>
+__device__ inline int swz_k(int row, int k) {
+  // SWZV=1 folds row bit 3 into the hash (see gemm_fp8_mx.hip: it fixed
+  // the fp8 sibling's r/r+8 b128-group collisions, +11% there)
+  int g = (row & 7) ^ (SWZV ? ((row >> 3) & 1) : 0);
+  return k ^ (g << 3);
+}
 
 // Stage one [128][64] half-tile: 1024 16-B slots, 512 threads x 2 passes.
 // LDS destination is lane-linear (slot order == lds address order); the
 // swizzle permutes which global 16-B chunk lands in each slot.
+template <int SWZV = 0>
 __device__ inline void stage_half(const __hip_bfloat16* gbase, int ldk,
                                   __hip_bfloat16* half_base) {
   const int t = threadIdx.x;
@@ -107,15 +114,16 @@ __device__ inline void stage_half(const __hip_bfloat16* gbase, int ldk,
     int row_l = slot >> 3;
     int chunk = (slot & 7) * 8;
     int src_row = swz_row(row_l);
-    glds16(gbase + src_row * ldk + swz_k(src_row, chunk),
+    glds16(gbase + src_row * ldk + swz_k<SWZV>(src_row, chunk),
            half_base + row_l * BK + chunk);
   }
 }
 
+template <int SWZV = 0>
 __device__ inline bf16x8 read_frag(const __hip_bfloat16* half_base, int row,
                                    int kbase) {
   return *reinterpret_cast<const bf16x8*>(half_base + swz_row(row) * BK +
-                                          swz_k(row, kbase));
+                                          swz_k<SWZV>(row, kbase));
 }
 
 __device__ inline void wait_lgkm0_fence() {
@@ -167,7 +175,7 @@ __device__ inline void wait_half_landed(int staged_through, int target) {
 // indices. SBASE = (4t)&7 — 0 for even tiles, 4 for odd ones — makes every
 // slot and staging destination a constant. LGKM selects the manual
 // lgkmcnt fences (PIPE=2) vs compiler-managed waits (PIPE=3).
-template <int SBASE, bool LGKM, typename STAGE>
+template <int SBASE, bool LGKM, int SWZV, typename STAGE>
 __device__ inline void tile4(const __hip_bfloat16* lds_c, STAGE&& stage_hs,
                              int t, int T, int arow, int brow, int fk,
                              bf16x8 (&afrag)[4][2], bf16x8 (&bfrag)[2][2][2],
@@ -182,12 +190,12 @@ __device__ inline void tile4(const __hip_bfloat16* lds_c, STAGE&& stage_hs,
   for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
-      afrag[fm][kk] = read_frag(As0, arow + fm * 16, kk * 32 + fk);
+      afrag[fm][kk] = read_frag<SWZV>(As0, arow + fm * 16, kk * 32 + fk);
 #pragma unroll
   for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
-      bfrag[0][fn][kk] = read_frag(Bs0, brow + fn * 16, kk * 32 + fk);
+      bfrag[0][fn][kk] = read_frag<SWZV>(Bs0, brow + fn * 16, kk * 32 + fk);
   if (4 * t + 7 < 4 * T) stage_hs(4 * t + 7, (SBASE + 7) & 7);
   if (LGKM) asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
   __builtin_amdgcn_s_barrier();
@@ -209,7 +217,7 @@ __device__ inline void tile4(const __hip_bfloat16* lds_c, STAGE&& stage_hs,
   for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
-      bfrag[1][fn][kk] = read_frag(Bs1, brow + fn * 16, kk * 32 + fk);
+      bfrag[1][fn][kk] = read_frag<SWZV>(Bs1, brow + fn * 16, kk * 32 + fk);
   if (4 * t + 8 < 4 * T) stage_hs(4 * t + 8, (SBASE + 8) & 7);
   __builtin_amdgcn_s_barrier();
   if (LGKM) wait_lgkm0_fence();
@@ -230,7 +238,7 @@ __device__ inline void tile4(const __hip_bfloat16* lds_c, STAGE&& stage_hs,
   for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
-      afrag[fm][kk] = read_frag(As1, arow + fm * 16, kk * 32 + fk);
+      afrag[fm][kk] = read_frag<SWZV>(As1, arow + fm * 16, kk * 32 + fk);
   if (4 * t + 9 < 4 * T) stage_hs(4 * t + 9, (SBASE + 9) & 7);
   __builtin_amdgcn_s_barrier();
   if (LGKM) wait_lgkm0_fence();
@@ -268,7 +276,7 @@ __device__ inline void tile4(const __hip_bfloat16* lds_c, STAGE&& stage_hs,
   __builtin_amdgcn_s_barrier();
 }
 
-template <int XCD, int PIPE>
+template <int XCD, int PIPE, int SWZV = 0>
 __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
     const __hip_bfloat16* __restrict__ A,   // [M][K]
     const __hip_bfloat16* __restrict__ Bt,  // [N][K]
@@ -317,7 +325,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
       g = Ablk + (long)128 * K + tt * BK;
     else
       g = Bblk + (long)128 * K + tt * BK;
-    stage_half(g, K, lds + (h & 7) * HALF_ELEMS);
+    stage_half<SWZV>(g, K, lds + (h & 7) * HALF_ELEMS);
   };
 
   const int arow = wm * 64 + frow;
@@ -345,16 +353,16 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
         g = Ablk + (long)128 * K + tt * BK;
       else
         g = Bblk + (long)128 * K + tt * BK;
-      stage_half(g, K, lds + slot * HALF_ELEMS);
+      stage_half<SWZV>(g, K, lds + slot * HALF_ELEMS);
     };
     constexpr bool LG = (PIPE == 2);
     int t = 0;
     for (; t + 1 < T; t += 2) {
-      tile4<0, LG>(lds, stage_hs, t, T, arow, brow, fk, afrag, bfrag, acc);
-      tile4<4, LG>(lds, stage_hs, t + 1, T, arow, brow, fk, afrag, bfrag, acc);
+      tile4<0, LG, SWZV>(lds, stage_hs, t, T, arow, brow, fk, afrag, bfrag, acc);
+      tile4<4, LG, SWZV>(lds, stage_hs, t + 1, T, arow, brow, fk, afrag, bfrag, acc);
     }
     if (t < T)  // odd T tail (t even here, slot base 0)
-      tile4<0, LG>(lds, stage_hs, t, T, arow, brow, fk, afrag, bfrag, acc);
+      tile4<0, LG, SWZV>(lds, stage_hs, t, T, arow, brow, fk, afrag, bfrag, acc);
   } else if (PIPE == 0 || PIPE == 4 || PIPE == 5) {
     // ---- round-1 schedule (+PIPE 4/5 micro-variants) ---------------------
     // prologue: tile 0 fully + 3 halves of tile 1 (guide's 4 then +3)
@@ -387,13 +395,13 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
       for (int fm = 0; fm < 4; ++fm) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
-          afrag[fm][kk] = read_frag(As0, arow + fm * 16, kk * 32 + fk);
+          afrag[fm][kk] = read_frag<SWZV>(As0, arow + fm * 16, kk * 32 + fk);
       }
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
-          bfrag[0][fn][kk] = read_frag(Bs0, brow + fn * 16, kk * 32 + fk);
+          bfrag[0][fn][kk] = read_frag<SWZV>(Bs0, brow + fn * 16, kk * 32 + fk);
       }
       if (PIPE == 4 && t > 0 && 4 * t + 6 < 4 * T)
         stage_h(4 * t + 6);  // deferred from the previous merged phase
@@ -420,7 +428,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
       for (int fn = 0; fn < 2; ++fn) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
-          bfrag[1][fn][kk] = read_frag(Bs1, brow + fn * 16, kk * 32 + fk);
+          bfrag[1][fn][kk] = read_frag<SWZV>(Bs1, brow + fn * 16, kk * 32 + fk);
       }
       if (4 * t + 8 < 4 * T) stage_h(4 * t + 8);
       __builtin_amdgcn_s_barrier();
@@ -442,7 +450,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
       for (int fm = 0; fm < 4; ++fm) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
-          afrag[fm][kk] = read_frag(As1, arow + fm * 16, kk * 32 + fk);
+          afrag[fm][kk] = read_frag<SWZV>(As1, arow + fm * 16, kk * 32 + fk);
       }
       if (4 * t + 9 < 4 * T) stage_h(4 * t + 9);
       __builtin_amdgcn_s_barrier();
@@ -512,7 +520,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
     for (int fm = 0; fm < 4; ++fm) {
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk)
-        a0frag[fm][kk] = read_frag(lds + 0 * HALF_ELEMS, arow + fm * 16,
+        a0frag[fm][kk] = read_frag<SWZV>(lds + 0 * HALF_ELEMS, arow + fm * 16,
                                    kk * 32 + fk);
     }
 
@@ -527,7 +535,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
       for (int fn = 0; fn < 2; ++fn) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
-          bfrag[0][fn][kk] = read_frag(Bs0, brow + fn * 16, kk * 32 + fk);
+          bfrag[0][fn][kk] = read_frag<SWZV>(Bs0, brow + fn * 16, kk * 32 + fk);
       }
       if (4 * t + 7 < 4 * T) stage_h(4 * t + 7);
       __builtin_amdgcn_s_barrier();
@@ -551,7 +559,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
       for (int fn = 0; fn < 2; ++fn) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
-          bfrag[1][fn][kk] = read_frag(Bs1, brow + fn * 16, kk * 32 + fk);
+          bfrag[1][fn][kk] = read_frag<SWZV>(Bs1, brow + fn * 16, kk * 32 + fk);
       }
       if (4 * t + 8 < 4 * T) stage_h(4 * t + 8);
       __builtin_amdgcn_s_barrier();
@@ -573,7 +581,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
       for (int fm = 0; fm < 4; ++fm) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
-          a1frag[fm][kk] = read_frag(As1, arow + fm * 16, kk * 32 + fk);
+          a1frag[fm][kk] = read_frag<SWZV>(As1, arow + fm * 16, kk * 32 + fk);
       }
       if (4 * t + 9 < 4 * T) stage_h(4 * t + 9);
       __builtin_amdgcn_s_barrier();
@@ -599,7 +607,7 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_8phase_kernel(
         for (int fm = 0; fm < 4; ++fm) {
 #pragma unroll
           for (int kk = 0; kk < 2; ++kk)
-            a0frag[fm][kk] = read_frag(As0n, arow + fm * 16, kk * 32 + fk);
+            a0frag[fm][kk] = read_frag<SWZV>(As0n, arow + fm * 16, kk * 32 + fk);
         }
       }
       if (4 * t + 10 < 4 * T) stage_h(4 * t + 10);

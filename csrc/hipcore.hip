@@ -578,6 +578,10 @@ void launch_8ph(int variant, dim3 grid, hipStream_t s,
       hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<1, 5>), grid,
                          dim3(512), 0, s, A, Bt, C, M, N, K);
       break;
+    case 12:  // PIPE=0 schedule + row-bit-3 LDS swizzle
+      hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<0, 0, 1>), grid,
+                         dim3(512), 0, s, A, Bt, C, M, N, K);
+      break;
     default:
       hipLaunchKernelGGL((gemm_bf16_8ph::gemm_bf16_8phase_kernel<0, 0>), grid,
                          dim3(512), 0, s, A, Bt, C, M, N, K);
