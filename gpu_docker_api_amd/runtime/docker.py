@@ -61,7 +61,10 @@ class DockerRuntime(RuntimeDriver):
                 data = await resp.read()
                 if resp.status == 404:
                     raise ContainerNotExist(path)
-                if resp.status == 409:
+                if resp.status == 409 and path.endswith("/create"):
+                    # name conflict on create; elsewhere 409 means a state
+                    # conflict (e.g. exec on a stopped container) and falls
+                    # through to the generic error below
                     raise ContainerExisted(path)
                 if resp.status not in ok:
                     msg = data.decode(errors="replace")

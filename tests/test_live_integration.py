@@ -377,3 +377,38 @@ def test_workqueue_survives_etcd_outage(tmp_path, run):
                     p.wait(timeout=5)
                 except Exception:
                     pass
+
+
+def test_dockerd_sim_error_paths(dockerd_sim, tmp_path, run):
+    """Engine-API error semantics over the socket: 409 on duplicate
+    create, 404 on missing containers/volumes, 409 on exec in a stopped
+    container — mapped to the driver's typed exceptions."""
+    from gpu_docker_api_amd.models.etcd import ContainerSpec
+    from gpu_docker_api_amd.runtime.docker import DockerRuntime
+    from gpu_docker_api_amd.xerrors import ContainerExisted, ContainerNotExist
+
+    async def main():
+        rt = DockerRuntime(socket_path=dockerd_sim)
+        spec = ContainerSpec()
+        spec.container_name = "e-1"
+        spec.config = {"Image": "img", "Cmd": ["sleep", "30"]}
+        await rt.create(spec)
+        with pytest.raises(ContainerExisted):
+            await rt.create(spec)
+        assert await rt.inspect("nope-1") is None
+        with pytest.raises(ContainerNotExist):
+            await rt.start("nope-1")
+        # exec against a created-but-not-started container: engine refuses
+        with pytest.raises((ContainerNotExist, RuntimeError)):
+            await rt.execute_rc("e-1", ["true"])
+        # unsafe name refused at the engine boundary (400 -> RuntimeError)
+        bad = ContainerSpec()
+        bad.container_name = "../esc"
+        bad.config = {"Image": "img"}
+        with pytest.raises(Exception):
+            await rt.create(bad)
+        await rt.remove("e-1", force=True)
+        assert await rt.volume_inspect("novol") is None
+        await rt.close()
+
+    run(main())

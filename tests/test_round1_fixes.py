@@ -176,3 +176,50 @@ def test_failed_replace_restores_spec_and_rollback_alignment(tmp_path, run):
         await d.stop()
 
     run(main())
+
+
+# ------------------------------------------------------- property fuzzing
+def test_safe_subpath_property_fuzz(tmp_path):
+    """For ANY string (hypothesis), safe_subpath either raises ValueError or
+    returns a path strictly below base after symlink/dot resolution."""
+    import os
+
+    from hypothesis import given, settings, strategies as st
+
+    from gpu_docker_api_amd.utils.names import safe_subpath
+
+    base = str(tmp_path / "base")
+    os.makedirs(base, exist_ok=True)
+    base_real = os.path.realpath(base)
+
+    @settings(max_examples=300, deadline=None)
+    @given(st.text(min_size=0, max_size=64))
+    def check(name):
+        try:
+            p = safe_subpath(base, name)
+        except ValueError:
+            return
+        assert os.path.realpath(p).startswith(base_real + os.sep)
+
+    check()
+
+
+def test_valid_name_implies_safe_subpath(tmp_path):
+    """Every name the routers accept must be safe to join."""
+    import os
+
+    from hypothesis import given, settings, strategies as st
+
+    from gpu_docker_api_amd.utils.names import safe_subpath, valid_name
+
+    base = str(tmp_path / "b")
+    os.makedirs(base, exist_ok=True)
+
+    @settings(max_examples=300, deadline=None)
+    @given(st.text(min_size=1, max_size=64))
+    def check(name):
+        if valid_name(name):
+            p = safe_subpath(base, name)  # must not raise
+            assert os.path.basename(p) == name
+
+    check()
