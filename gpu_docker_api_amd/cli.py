@@ -213,7 +213,7 @@ def res_validate(
     size: int = typer.Option(4096, "--size"),
     iters: int = typer.Option(5, "--iters"),
 ):
-    """Burn-in free GPUs (HBM bandwidth + bf16 MFMA TFLOPS)."""
+    """Burn-in free GPUs (HBM bandwidth + bf16/MX-fp8/MX-fp4 MFMA TFLOPS)."""
     with _client() as c:
         _show(
             c.post(
