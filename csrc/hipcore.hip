@@ -691,6 +691,12 @@ torch::Tensor gemm_fp8_mx_bt(torch::Tensor A, torch::Tensor Bt, int shape) {
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
                        C.data_ptr<float>(), M, N, K);
+  else if (shape == 18)  // + quad-transpose dwordx4 epilogue
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1, 1>), dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
   else if (shape == 17)  // legacy row&7-only swizzle (A/B reference)
     hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 0>), dim3(grid), dim3(512),
                        0, stream.stream(),
@@ -723,6 +729,12 @@ torch::Tensor gemm_fp4_mx_bt(torch::Tensor A, torch::Tensor Bt, int K, int shape
   auto stream = at::hip::getCurrentHIPStream();
   if (shape == 32)
     hipLaunchKernelGGL((gemm_fp8_mx::gemm_mx32_kernel<4, 0>), dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
+  else if (shape == 18)  // + quad-transpose dwordx4 epilogue
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0, 1>), dim3(grid), dim3(512),
                        0, stream.stream(),
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
@@ -766,6 +778,9 @@ double gemm_fp8_mx_tflops(int device, int size, int iters, int shape) {
         else if (shape == 17)  // legacy swizzle (A/B reference)
           hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 0>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
+        else if (shape == 18)  // + quad-transpose dwordx4 epilogue
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1, 1>), dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
         else  // default: row-bit-3 swizzle
           hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
@@ -800,6 +815,9 @@ double gemm_fp4_mx_tflops(int device, int size, int iters, int shape) {
                              dim3(512), 0, s, A, Bt, C, size, size, size);
         else if (shape == 17)
           hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 1>), dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
+        else if (shape == 18)  // + quad-transpose dwordx4 epilogue
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0, 1>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
         else
           hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0>), dim3(grid),
