@@ -285,9 +285,11 @@ class ProcRuntime(RuntimeDriver):
         if p.state.running:
             return
         cmd = list(p.spec.config.get("Cmd") or []) or DEFAULT_CMD
-        # spawn inline: fork+exec is ~1 ms, and forking from an executor
-        # thread measurably REGRESSES multi-tenant throughput (fork while
-        # pool threads hold allocator locks) — measured 109 -> 71 cycles/s
+        # spawn inline: fork+exec is ~0.7 ms; offloading it to an executor
+        # (shared pool in round 1: 109 -> 71 cycles/s; a DEDICATED 2-thread
+        # pool re-measured in round 2: 256 -> 211 single-tenant, 258 -> 243
+        # at 8 tenants, bench p50 4.5 -> 5.0 ms) loses more to the thread
+        # hop than the freed loop time buys — keep it inline
         logf = open(os.path.join(self._cdir(name), "console.log"), "ab")
         try:
             p.popen = subprocess.Popen(
