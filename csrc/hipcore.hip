@@ -691,6 +691,12 @@ torch::Tensor gemm_fp8_mx_bt(torch::Tensor A, torch::Tensor Bt, int shape) {
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
                        C.data_ptr<float>(), M, N, K);
+  else if (shape == 20)  // merged phases 2+3
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1, 0, 1>), dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
   else if (shape == 18)  // + quad-transpose dwordx4 epilogue
     hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1, 1>), dim3(grid), dim3(512),
                        0, stream.stream(),
@@ -704,7 +710,7 @@ torch::Tensor gemm_fp8_mx_bt(torch::Tensor A, torch::Tensor Bt, int shape) {
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
                        C.data_ptr<float>(), M, N, K);
   else  // default: row-bit-3 swizzle (conflict-free for the 2-chunk reads, +11%)
-    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1>), dim3(grid), dim3(512),
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1, 0, 0>), dim3(grid), dim3(512),
                        0, stream.stream(),
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
@@ -733,6 +739,12 @@ torch::Tensor gemm_fp4_mx_bt(torch::Tensor A, torch::Tensor Bt, int K, int shape
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
                        C.data_ptr<float>(), M, N, K);
+  else if (shape == 19)  // merged phases 2+3
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0, 0, 1>), dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
   else if (shape == 18)  // + quad-transpose dwordx4 epilogue
     hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0, 1>), dim3(grid), dim3(512),
                        0, stream.stream(),
@@ -745,8 +757,14 @@ torch::Tensor gemm_fp4_mx_bt(torch::Tensor A, torch::Tensor Bt, int K, int shape
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
                        C.data_ptr<float>(), M, N, K);
-  else
-    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0>), dim3(grid), dim3(512),
+  else if (shape == 21)  // unmerged legacy (A/B reference)
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0, 0, 0>), dim3(grid), dim3(512),
+                       0, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(A.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
+                       C.data_ptr<float>(), M, N, K);
+  else  // default: merged phases 2+3 (+1-2% measured both orders)
+    hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0, 0, 1>), dim3(grid), dim3(512),
                        0, stream.stream(),
                        reinterpret_cast<const unsigned char*>(A.data_ptr()),
                        reinterpret_cast<const unsigned char*>(Bt.data_ptr()),
@@ -777,6 +795,9 @@ double gemm_fp8_mx_tflops(int device, int size, int iters, int shape) {
                              dim3(512), 0, s, A, Bt, C, size, size, size);
         else if (shape == 17)  // legacy swizzle (A/B reference)
           hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 0>), dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
+        else if (shape == 20)  // merged phases 2+3
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1, 0, 1>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
         else if (shape == 18)  // + quad-transpose dwordx4 epilogue
           hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<0, 1, 1>), dim3(grid),
@@ -816,11 +837,17 @@ double gemm_fp4_mx_tflops(int device, int size, int iters, int shape) {
         else if (shape == 17)
           hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 1>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
+        else if (shape == 19)  // merged phases 2+3
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0, 0, 1>), dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
         else if (shape == 18)  // + quad-transpose dwordx4 epilogue
           hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0, 1>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
-        else
-          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0>), dim3(grid),
+        else if (shape == 21)  // unmerged legacy
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0, 0, 0>), dim3(grid),
+                             dim3(512), 0, s, A, Bt, C, size, size, size);
+        else  // default: merged phases 2+3
+          hipLaunchKernelGGL((gemm_fp8_mx::gemm_fp8_mx_kernel<4, 0, 0, 1>), dim3(grid),
                              dim3(512), 0, s, A, Bt, C, size, size, size);
       },
       iters);
