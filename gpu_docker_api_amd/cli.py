@@ -124,6 +124,18 @@ def exec(name: str, command: str, workdir: str = typer.Option("", "--workdir", "
 
 
 @app.command()
+def logs(name: str, tail: int = typer.Option(200, "--tail", "-n")):
+    """Captured console output of the current version."""
+    with _client() as c:
+        r = c.get(f"/api/v1/replicaSet/{name}/logs", params={"tail": str(tail)})
+        body = r.json()
+        if body.get("code") == 200:
+            typer.echo((body.get("data") or {}).get("logs", ""), nl=False)
+        else:
+            _show(r)
+
+
+@app.command()
 def commit(name: str, image: str):
     with _client() as c:
         _show(c.post(f"/api/v1/replicaSet/{name}/commit", json={"newImageName": image}))

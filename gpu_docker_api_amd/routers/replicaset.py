@@ -96,6 +96,22 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
             return error(map_error(exc, Code.CONTAINER_GET_INFO_FAILED))
         return success(data)
 
+    @r.get("/{name}/logs")
+    async def logs(name: str, request: Request):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
+        try:
+            tail = min(10000, max(1, int(request.query_params.get("tail", "200"))))
+        except ValueError:
+            return error(Code.INVALID_PARAMS)
+        try:
+            data = await svc.get_container_logs(name, tail=tail)
+        except Exception as exc:  # noqa: BLE001
+            log_error("replicaSet.logs", exc)
+            return error(map_error(exc, Code.CONTAINER_GET_INFO_FAILED))
+        return success({"logs": data})
+
     @r.get("/{name}/history")
     async def history(name: str):
         bad = _bad_name(name)

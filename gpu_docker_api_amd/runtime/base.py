@@ -103,6 +103,11 @@ class RuntimeDriver:
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         raise NotImplementedError
 
+    async def logs(self, name: str, tail: int = 200) -> str:
+        """Captured container output (extension: the reference exposes no
+        logs route; its users must go to dockerd directly)."""
+        raise NotImplementedError
+
     # ---- volumes ----
     async def volume_create(
         self, name: str, driver_opts: Optional[Dict[str, str]] = None

@@ -229,6 +229,15 @@ class DockerRuntime(RuntimeDriver):
             rc = 0
         return self._demux_stream(raw or b""), rc
 
+    async def logs(self, name: str, tail: int = 200) -> str:
+        raw = await self._call(
+            "GET",
+            f"/containers/{name}/logs",
+            params={"stdout": "true", "stderr": "true", "tail": str(tail)},
+            raw=True,
+        )
+        return self._demux_stream(raw or b"")
+
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         repo, _, t = image.partition(":")
         if not t:

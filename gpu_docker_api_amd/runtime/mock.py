@@ -32,6 +32,7 @@ class MockRuntime(RuntimeDriver):
         self.containers: Dict[str, ContainerState] = {}
         self.volumes: Dict[str, VolumeState] = {}
         self.images: Dict[str, str] = {}  # image name -> seed dir (optional)
+        self._logs: Dict[str, str] = {}   # synthetic console output
         # test hook: operation names that should raise (failure-injection)
         self.fail_on = fail_on or set()
 
@@ -127,6 +128,11 @@ class MockRuntime(RuntimeDriver):
         )
         out, _ = await proc.communicate()
         return out.decode(errors="replace"), proc.returncode
+
+    async def logs(self, name: str, tail: int = 200) -> str:
+        if name not in self.containers:
+            raise ContainerNotExist(name)
+        return self._logs.get(name, "")
 
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         self._maybe_fail("commit")

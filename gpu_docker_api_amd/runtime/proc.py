@@ -444,6 +444,21 @@ class ProcRuntime(RuntimeDriver):
         out, _ = await proc.communicate()
         return out.decode(errors="replace"), proc.returncode
 
+    async def logs(self, name: str, tail: int = 200) -> str:
+        p = self._get(name)
+        path = os.path.join(self._cdir(p.state.name), "console.log")
+        try:
+            with open(path, "rb") as f:
+                f.seek(0, os.SEEK_END)
+                size = f.tell()
+                # cap the read at the last MiB; tail-line slicing follows
+                f.seek(max(0, size - (1 << 20)))
+                data = f.read()
+        except OSError:
+            return ""
+        lines = data.decode(errors="replace").splitlines()
+        return "\n".join(lines[-tail:]) + ("\n" if lines else "")
+
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         p = self._get(name)
         ref = f"{image}:{tag}" if tag else image

@@ -618,6 +618,12 @@ class ReplicaSetService:
             raise ContainerNotExist(name)
         return ContainerSpec.deserialize(kv.value).to_dict()
 
+    async def get_container_logs(self, name: str, tail: int = 200) -> str:
+        """GET /{name}/logs (extension): captured console output of the
+        current version. The reference exposes no logs route — its users
+        must query dockerd directly."""
+        return await self.runtime.logs(self._current_vname(name), tail=tail)
+
     async def get_container_history(self, name: str) -> List[Dict]:
         """GET /{name}/history — all surviving versions, newest first
         (reference GetContainerHistory, replicaset.go:908-929, via the

@@ -187,6 +187,17 @@ class DockerdSim:
             return self._err(404, "no such exec")
         return web.json_response({"ExitCode": e["exit_code"] or 0, "Running": False})
 
+    async def container_logs(self, request: web.Request) -> web.Response:
+        name = request.match_info["name"]
+        tail = int(request.query.get("tail", "200"))
+        try:
+            out = await self.rt.logs(name, tail=tail)
+        except ContainerNotExist:
+            return self._err(404, f"No such container: {name}")
+        payload = out.encode()
+        frame = bytes([1, 0, 0, 0]) + struct.pack(">I", len(payload)) + payload
+        return web.Response(body=frame, content_type="application/vnd.docker.raw-stream")
+
     async def commit(self, request: web.Request) -> web.Response:
         name = request.query.get("container", "")
         repo = request.query.get("repo", "")
@@ -235,6 +246,7 @@ class DockerdSim:
         r.add_post("/{v}/containers/create", self.create)
         r.add_get("/{v}/containers/json", self.list_)
         r.add_get("/{v}/containers/{name}/json", self.inspect)
+        r.add_get("/{v}/containers/{name}/logs", self.container_logs)
         r.add_post("/{v}/containers/{name}/exec", self.exec_create)
         r.add_post("/{v}/containers/{name}/{verb}", self.lifecycle)
         r.add_delete("/{v}/containers/{name}", self.remove)

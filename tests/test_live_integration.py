@@ -412,3 +412,27 @@ def test_dockerd_sim_error_paths(dockerd_sim, tmp_path, run):
         await rt.close()
 
     run(main())
+
+
+def test_docker_logs_over_socket(dockerd_sim, tmp_path, run):
+    from gpu_docker_api_amd.models.etcd import ContainerSpec
+    from gpu_docker_api_amd.runtime.docker import DockerRuntime
+
+    async def main():
+        rt = DockerRuntime(socket_path=dockerd_sim)
+        spec = ContainerSpec()
+        spec.container_name = "lg-1"
+        spec.config = {"Image": "img", "Cmd": ["sh", "-c", "echo over-the-wire; sleep 30"]}
+        await rt.create(spec)
+        await rt.start("lg-1")
+        out = ""
+        for _ in range(100):
+            out = await rt.logs("lg-1")
+            if "over-the-wire" in out:
+                break
+            await asyncio.sleep(0.05)
+        assert "over-the-wire" in out
+        await rt.remove("lg-1", force=True)
+        await rt.close()
+
+    run(main())

@@ -429,3 +429,41 @@ def test_container_patch_volume_bind(tmp_path, run):
         await d.stop()
 
     run(main())
+
+
+def test_container_logs_endpoint(tmp_path, run):
+    """Logs extension: the proc runtime captures console output and the
+    service/route surface it with tail semantics."""
+    import asyncio
+
+    from gpu_docker_api_amd.models import ContainerRun
+
+    async def main():
+        from helpers import make_config
+        from gpu_docker_api_amd.routers.app import Daemon
+
+        cfg = make_config(tmp_path, runtime="proc")
+        d = Daemon(cfg)
+        await d.start()
+        await d.replicaset.run_gpu_container(
+            ContainerRun(
+                image_name="img",
+                replica_set_name="logs",
+                gpu_count=0,
+                cmd=["sh", "-c", "echo line-one; echo line-two; sleep 30"],
+            )
+        )
+        out = ""
+        for _ in range(100):  # console.log is written asynchronously
+            out = await d.replicaset.get_container_logs("logs")
+            if "line-two" in out:
+                break
+            await asyncio.sleep(0.05)
+        assert "line-one" in out and "line-two" in out
+        # tail=1 returns only the newest line
+        tail1 = await d.replicaset.get_container_logs("logs", tail=1)
+        assert "line-two" in tail1 and "line-one" not in tail1
+        await d.replicaset.delete_container("logs")
+        await d.stop()
+
+    run(main())
