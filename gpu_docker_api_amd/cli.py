@@ -136,6 +136,13 @@ def logs(name: str, tail: int = typer.Option(200, "--tail", "-n")):
 
 
 @app.command()
+def stats(name: str):
+    """Live cpu/memory/pids of the current version."""
+    with _client() as c:
+        _show(c.get(f"/api/v1/replicaSet/{name}/stats"))
+
+
+@app.command()
 def commit(name: str, image: str):
     with _client() as c:
         _show(c.post(f"/api/v1/replicaSet/{name}/commit", json={"newImageName": image}))

@@ -112,6 +112,18 @@ def make_router(svc: ReplicaSetService) -> APIRouter:
             return error(map_error(exc, Code.CONTAINER_GET_INFO_FAILED))
         return success({"logs": data})
 
+    @r.get("/{name}/stats")
+    async def stats(name: str):
+        bad = _bad_name(name)
+        if bad is not None:
+            return bad
+        try:
+            data = await svc.get_container_stats(name)
+        except Exception as exc:  # noqa: BLE001
+            log_error("replicaSet.stats", exc)
+            return error(map_error(exc, Code.CONTAINER_GET_INFO_FAILED))
+        return success(data)
+
     @r.get("/{name}/history")
     async def history(name: str):
         bad = _bad_name(name)

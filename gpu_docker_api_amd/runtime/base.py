@@ -108,6 +108,11 @@ class RuntimeDriver:
         logs route; its users must go to dockerd directly)."""
         raise NotImplementedError
 
+    async def stats(self, name: str) -> Dict:
+        """Live resource usage {cpuSeconds, memoryBytes, pids} (extension:
+        the reference has no stats route)."""
+        raise NotImplementedError
+
     # ---- volumes ----
     async def volume_create(
         self, name: str, driver_opts: Optional[Dict[str, str]] = None

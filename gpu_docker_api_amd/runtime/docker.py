@@ -238,6 +238,21 @@ class DockerRuntime(RuntimeDriver):
         )
         return self._demux_stream(raw or b"")
 
+    async def stats(self, name: str) -> Dict[str, Any]:
+        d = await self._call(
+            "GET", f"/containers/{name}/stats",
+            params={"stream": "false", "one-shot": "true"},
+        )
+        cpu_ns = ((d.get("cpu_stats") or {}).get("cpu_usage") or {}).get("total_usage", 0)
+        mem = (d.get("memory_stats") or {}).get("usage", 0)
+        pids = ((d.get("pids_stats") or {}).get("current")) or 0
+        return {
+            "running": True,
+            "cpuSeconds": round(int(cpu_ns) / 1e9, 3),
+            "memoryBytes": int(mem),
+            "pids": int(pids),
+        }
+
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         repo, _, t = image.partition(":")
         if not t:

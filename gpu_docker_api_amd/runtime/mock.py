@@ -134,6 +134,12 @@ class MockRuntime(RuntimeDriver):
             raise ContainerNotExist(name)
         return self._logs.get(name, "")
 
+    async def stats(self, name: str) -> dict:
+        if name not in self.containers:
+            raise ContainerNotExist(name)
+        st = self.containers[name]
+        return {"running": st.running, "cpuSeconds": 0.0, "memoryBytes": 0, "pids": 0}
+
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         self._maybe_fail("commit")
         st = self._get(name)

@@ -444,6 +444,45 @@ class ProcRuntime(RuntimeDriver):
         out, _ = await proc.communicate()
         return out.decode(errors="replace"), proc.returncode
 
+    async def stats(self, name: str) -> Dict:
+        """Process-group resource usage: cgroup v2 figures when the group
+        has one (exact), else /proc of the leader process."""
+        p = self._get(name)
+        self._refresh(p)
+        out: Dict = {"running": p.state.running, "cpuSeconds": 0.0,
+                     "memoryBytes": 0, "pids": 0}
+        if not p.state.running or p.popen is None:
+            return out
+        if p.cgroup:
+            try:
+                with open(os.path.join(p.cgroup, "cpu.stat")) as f:
+                    for line in f:
+                        if line.startswith("usage_usec"):
+                            out["cpuSeconds"] = round(int(line.split()[1]) / 1e6, 3)
+                with open(os.path.join(p.cgroup, "memory.current")) as f:
+                    out["memoryBytes"] = int(f.read())
+                with open(os.path.join(p.cgroup, "cgroup.procs")) as f:
+                    out["pids"] = len(f.read().split())
+                return out
+            except OSError:
+                pass
+        try:
+            pid = p.popen.pid
+            with open(f"/proc/{pid}/stat") as f:
+                fields = f.read().rsplit(") ", 1)[1].split()
+                # utime (11) + stime (12) after the comm field, in ticks
+                ticks = int(fields[11]) + int(fields[12])
+                out["cpuSeconds"] = round(ticks / os.sysconf("SC_CLK_TCK"), 3)
+            with open(f"/proc/{pid}/status") as f:
+                for line in f:
+                    if line.startswith("VmRSS:"):
+                        out["memoryBytes"] = int(line.split()[1]) * 1024
+                        break
+            out["pids"] = 1
+        except (OSError, IndexError, ValueError):
+            pass
+        return out
+
     async def logs(self, name: str, tail: int = 200) -> str:
         p = self._get(name)
         path = os.path.join(self._cdir(p.state.name), "console.log")

@@ -624,6 +624,11 @@ class ReplicaSetService:
         must query dockerd directly."""
         return await self.runtime.logs(self._current_vname(name), tail=tail)
 
+    async def get_container_stats(self, name: str) -> Dict:
+        """GET /{name}/stats (extension): live cpu/memory/pids of the
+        current version."""
+        return await self.runtime.stats(self._current_vname(name))
+
     async def get_container_history(self, name: str) -> List[Dict]:
         """GET /{name}/history — all surviving versions, newest first
         (reference GetContainerHistory, replicaset.go:908-929, via the

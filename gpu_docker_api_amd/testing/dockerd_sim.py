@@ -198,6 +198,20 @@ class DockerdSim:
         frame = bytes([1, 0, 0, 0]) + struct.pack(">I", len(payload)) + payload
         return web.Response(body=frame, content_type="application/vnd.docker.raw-stream")
 
+    async def container_stats(self, request: web.Request) -> web.Response:
+        name = request.match_info["name"]
+        try:
+            st = await self.rt.stats(name)
+        except ContainerNotExist:
+            return self._err(404, f"No such container: {name}")
+        return web.json_response(
+            {
+                "cpu_stats": {"cpu_usage": {"total_usage": int(st["cpuSeconds"] * 1e9)}},
+                "memory_stats": {"usage": st["memoryBytes"]},
+                "pids_stats": {"current": st["pids"]},
+            }
+        )
+
     async def commit(self, request: web.Request) -> web.Response:
         name = request.query.get("container", "")
         repo = request.query.get("repo", "")
@@ -247,6 +261,7 @@ class DockerdSim:
         r.add_get("/{v}/containers/json", self.list_)
         r.add_get("/{v}/containers/{name}/json", self.inspect)
         r.add_get("/{v}/containers/{name}/logs", self.container_logs)
+        r.add_get("/{v}/containers/{name}/stats", self.container_stats)
         r.add_post("/{v}/containers/{name}/exec", self.exec_create)
         r.add_post("/{v}/containers/{name}/{verb}", self.lifecycle)
         r.add_delete("/{v}/containers/{name}", self.remove)

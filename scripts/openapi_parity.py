@@ -63,6 +63,7 @@ OUR_EXTENSION_FIELDS = {
 OUR_EXTENSION_ROUTES = {
     ("GET", "/api/v1/replicaSet"): "list all replicaSets (reference clients must track names externally)",
     ("GET", "/api/v1/replicaSet/{name}/logs"): "captured console output (reference users must query dockerd directly)",
+    ("GET", "/api/v1/replicaSet/{name}/stats"): "live cpu/memory/pids of the current version",
     ("GET", "/api/v1/volumes"): "list all volumes",
     ("GET", "/api/v1/resources/gpus/detail"): "per-GPU HBM bytes + measured xGMI adjacency (SURVEY §2.4 row 7)",
     ("POST", "/api/v1/resources/gpus/validate"): "MFMA burn-in validation of a GPU set before handing it to a tenant",

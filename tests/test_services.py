@@ -467,3 +467,31 @@ def test_container_logs_endpoint(tmp_path, run):
         await d.stop()
 
     run(main())
+
+
+def test_container_stats_endpoint(tmp_path, run):
+    """Stats extension: live cpu/memory of the current version (proc
+    runtime reads cgroup v2 or /proc)."""
+    import asyncio
+
+    from gpu_docker_api_amd.models import ContainerRun
+
+    async def main():
+        from helpers import make_config
+        from gpu_docker_api_amd.routers.app import Daemon
+
+        cfg = make_config(tmp_path, runtime="proc")
+        d = Daemon(cfg)
+        await d.start()
+        await d.replicaset.run_gpu_container(
+            ContainerRun(image_name="img", replica_set_name="st", gpu_count=0,
+                         cmd=["sleep", "30"])
+        )
+        st = await d.replicaset.get_container_stats("st")
+        assert st["running"] is True
+        assert st["memoryBytes"] > 0
+        assert st["pids"] >= 1
+        await d.replicaset.delete_container("st")
+        await d.stop()
+
+    run(main())
