@@ -626,8 +626,24 @@ class ReplicaSetService:
 
     async def get_container_stats(self, name: str) -> Dict:
         """GET /{name}/stats (extension): live cpu/memory/pids of the
-        current version."""
-        return await self.runtime.stats(self._current_vname(name))
+        current version, plus per-allocated-GPU HBM usage (288 GB HBM3E
+        per MI355X; the reference reports no per-container usage at all)."""
+        spec = await self._load_spec(name)
+        out = await self.runtime.stats(self._current_vname(name))
+        if spec.gpu_uuids:
+            try:
+                fresh = {g.uuid: g for g in (self.gpu.inventory.refresh_usage() or [])}
+            except Exception:  # inventory backend without live usage
+                fresh = {}
+            out["gpus"] = [
+                {
+                    "uuid": u,
+                    "hbmUsedBytes": fresh[u].vram_used if u in fresh else None,
+                    "hbmTotalBytes": fresh[u].vram_total if u in fresh else None,
+                }
+                for u in spec.gpu_uuids
+            ]
+        return out
 
     async def get_container_history(self, name: str) -> List[Dict]:
         """GET /{name}/history — all surviving versions, newest first

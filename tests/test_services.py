@@ -495,3 +495,24 @@ def test_container_stats_endpoint(tmp_path, run):
         await d.stop()
 
     run(main())
+
+
+def test_container_stats_includes_gpu_hbm(tmp_path, run):
+    from gpu_docker_api_amd.models import ContainerRun
+
+    async def main():
+        from helpers import make_daemon
+
+        d = await make_daemon(tmp_path)
+        await d.replicaset.run_gpu_container(
+            ContainerRun(image_name="img", replica_set_name="g", gpu_count=2)
+        )
+        st = await d.replicaset.get_container_stats("g")
+        assert len(st["gpus"]) == 2
+        for g in st["gpus"]:
+            assert g["uuid"].startswith("MockMI355X-")
+            assert g["hbmTotalBytes"] and g["hbmTotalBytes"] > 200 * 1024**3
+        await d.replicaset.delete_container("g")
+        await d.stop()
+
+    run(main())
