@@ -59,6 +59,11 @@ class _Proc:
         )
         self.manually_stopped = False
         self.restarts = 0
+        # daemon-restart reattachment: leader pid + its /proc starttime
+        # (detects pid reuse) for containers adopted without a Popen handle
+        self.attached_pid = 0
+        self.attached_start = 0
+        self.was_started = False
 
 
 class ProcRuntime(RuntimeDriver):
@@ -83,6 +88,13 @@ class ProcRuntime(RuntimeDriver):
         self.volumes: Dict[str, VolumeState] = {}
         self._supervisor: Optional[asyncio.Task] = None
         self._load_volumes()
+        self._load_containers()
+        try:  # adopt/resurrect across daemon restarts when a loop exists
+            asyncio.get_running_loop()
+            if self._procs:
+                self._ensure_supervisor()
+        except RuntimeError:
+            pass
 
     # -------------------------------------------------------------- supervisor
     def _ensure_supervisor(self) -> None:
@@ -104,7 +116,7 @@ class ProcRuntime(RuntimeDriver):
                     and p.state.status == "exited"
                     and not p.manually_stopped
                     and p.restart_policy in ("always", "unless-stopped")
-                    and p.popen is not None  # was started at least once
+                    and (p.popen is not None or p.was_started)
                 ):
                     try:
                         p.restarts += 1
@@ -165,11 +177,109 @@ class ProcRuntime(RuntimeDriver):
                             pass
                 self.volumes[name] = VolumeState(name=name, mountpoint=mp, options=opts)
 
+    @staticmethod
+    def _proc_starttime(pid: int) -> Optional[int]:
+        """starttime (clock ticks since boot) of a pid — stable identity
+        that survives pid reuse. Zombies (state Z — dead but unreaped by
+        their original parent, which may have been the previous daemon's
+        popen) count as gone."""
+        try:
+            with open(f"/proc/{pid}/stat") as f:
+                fields = f.read().rsplit(") ", 1)[1].split()
+                if fields[0] in ("Z", "X"):
+                    return None
+                return int(fields[19])
+        except (OSError, IndexError, ValueError):
+            return None
+
+    def _alive(self, p: _Proc) -> bool:
+        if p.popen is not None:
+            return p.popen.poll() is None
+        if p.attached_pid:
+            return self._proc_starttime(p.attached_pid) == p.attached_start
+        return False
+
+    def _leader_pid(self, p: _Proc) -> int:
+        if p.popen is not None:
+            return p.popen.pid
+        return p.attached_pid
+
     def _refresh(self, p: _Proc) -> None:
-        if p.popen is not None and p.popen.poll() is not None and p.state.running:
+        if p.state.running and not self._alive(p):
             p.state.running = False
             p.state.status = "exited"
             p.state.pid = 0
+            p.attached_pid = 0
+            self._write_meta(p)
+
+    def _write_meta(self, p: _Proc) -> None:
+        """Persist runtime identity next to spec.json so a daemon restart
+        can adopt still-running containers (dockerd containers survive the
+        daemon; round 1's proc runtime forgot everything)."""
+        try:
+            meta = {
+                "id": p.state.id,
+                "pid": self._leader_pid(p) if p.state.running else 0,
+                "starttime": (
+                    self._proc_starttime(self._leader_pid(p)) if p.state.running else 0
+                ),
+                "running": p.state.running,
+                "restarts": p.restarts,
+            }
+            with open(os.path.join(self._cdir(p.state.name), "meta.json"), "w") as f:
+                json.dump(meta, f)
+        except OSError:
+            pass
+
+    def _load_containers(self) -> None:
+        croot = os.path.join(self.base, "containers")
+        for name in sorted(os.listdir(croot)):
+            specf = os.path.join(croot, name, "spec.json")
+            if not os.path.isfile(specf):
+                continue
+            try:
+                spec = ContainerSpec.deserialize(open(specf).read())
+            except Exception:
+                continue
+            meta = {}
+            try:
+                meta = json.load(open(os.path.join(croot, name, "meta.json")))
+            except Exception:
+                pass
+            rootfs = os.path.join(croot, name, "rootfs")
+            st = ContainerState(
+                id=meta.get("id", ""),
+                name=name,
+                image=spec.image,
+                status="exited",
+                env=list(spec.env),
+                gpu_uuids=list(spec.gpu_uuids),
+                cpuset_cpus=spec.cpuset_cpus,
+                memory=spec.memory_bytes,
+                port_bindings=dict(spec.host_config.get("PortBindings") or {}),
+                upper_dir=rootfs,
+                binds=list(spec.host_config.get("Binds") or []),
+            )
+            p = _Proc(st, spec)
+            p.restarts = int(meta.get("restarts", 0) or 0)
+            pid = int(meta.get("pid", 0) or 0)
+            if meta.get("running") and pid:
+                if self._proc_starttime(pid) == meta.get("starttime"):
+                    # the workload outlived the daemon: adopt it live
+                    p.attached_pid = pid
+                    p.attached_start = int(meta["starttime"])
+                    p.was_started = True
+                    st.pid = pid
+                    st.running, st.status = True, "running"
+                else:
+                    # it died while unsupervised; the restart policy decides
+                    # whether the supervisor resurrects it (docker restarts
+                    # unless-stopped containers on daemon start)
+                    p.was_started = True
+                    p.manually_stopped = False
+            else:
+                p.manually_stopped = True
+            self._procs[name] = p
 
     # ------------------------------------------------------------ containers
     async def create(self, spec: ContainerSpec) -> str:
@@ -215,6 +325,7 @@ class ProcRuntime(RuntimeDriver):
         self._procs[name] = p
         with open(os.path.join(self._cdir(name), "spec.json"), "w") as f:
             f.write(spec.serialize())
+        self._write_meta(p)
         return cid
 
     def _get(self, name: str) -> _Proc:
@@ -302,15 +413,18 @@ class ProcRuntime(RuntimeDriver):
             )
         finally:
             logf.close()
+        p.attached_pid = 0
+        p.was_started = True
         p.state.pid = p.popen.pid
         p.state.running, p.state.paused, p.state.status = True, False, "running"
         self._setup_cgroup(name, p, p.popen.pid)
+        self._write_meta(p)
 
     def _signal_group(self, p: _Proc, sig: int) -> None:
-        if p.popen is None or p.popen.poll() is not None:
+        if not self._alive(p):
             return
         try:
-            os.killpg(os.getpgid(p.popen.pid), sig)
+            os.killpg(os.getpgid(self._leader_pid(p)), sig)
         except ProcessLookupError:
             pass
 
@@ -318,7 +432,7 @@ class ProcRuntime(RuntimeDriver):
         p = self._get(name)
         p.manually_stopped = True
         self._refresh(p)
-        if p.popen is not None and p.popen.poll() is None:
+        if self._alive(p):
             self._signal_group(p, signal.SIGTERM)
             # a paused (SIGSTOPped) group keeps SIGTERM pending forever:
             # continue it so termination can be delivered (docker semantics)
@@ -330,30 +444,33 @@ class ProcRuntime(RuntimeDriver):
             # and cooperative under concurrency (a blocking spin here
             # measurably cut 8-tenant throughput)
             spin_until = time.monotonic() + 0.002
-            while time.monotonic() < spin_until and p.popen.poll() is None:
+            while time.monotonic() < spin_until and self._alive(p):
                 await asyncio.sleep(0)
             deadline = time.monotonic() + timeout
             delay = 0.001  # timer-based backoff for the slow case
-            while time.monotonic() < deadline and p.popen.poll() is None:
+            while time.monotonic() < deadline and self._alive(p):
                 await asyncio.sleep(delay)
                 delay = min(delay * 2, 0.02)
-            if p.popen.poll() is None:
+            if self._alive(p):
                 self._signal_group(p, signal.SIGKILL)
                 # reap without blocking the event loop: a process stuck in
                 # D-state I/O would otherwise stall every handler for 5 s
                 # 
                 kill_deadline = time.monotonic() + 5
-                while time.monotonic() < kill_deadline and p.popen.poll() is None:
+                while time.monotonic() < kill_deadline and self._alive(p):
                     await asyncio.sleep(0.005)
-        if p.popen is not None:
+        leader = self._leader_pid(p)
+        if leader:
             # the leader is gone; sweep any group stragglers (orphaned
             # children). pgid == leader pid (start_new_session).
             try:
-                os.killpg(p.popen.pid, signal.SIGKILL)
+                os.killpg(leader, signal.SIGKILL)
             except (ProcessLookupError, PermissionError):
                 pass
         p.state.running, p.state.paused, p.state.status = False, False, "exited"
         p.state.pid = 0
+        p.attached_pid = 0
+        self._write_meta(p)
 
     async def pause(self, name: str) -> None:
         p = self._get(name)
@@ -451,7 +568,7 @@ class ProcRuntime(RuntimeDriver):
         self._refresh(p)
         out: Dict = {"running": p.state.running, "cpuSeconds": 0.0,
                      "memoryBytes": 0, "pids": 0}
-        if not p.state.running or p.popen is None:
+        if not p.state.running or not self._leader_pid(p):
             return out
         if p.cgroup:
             try:
@@ -467,7 +584,7 @@ class ProcRuntime(RuntimeDriver):
             except OSError:
                 pass
         try:
-            pid = p.popen.pid
+            pid = self._leader_pid(p)
             with open(f"/proc/{pid}/stat") as f:
                 fields = f.read().rsplit(") ", 1)[1].split()
                 # utime (11) + stime (12) after the comm field, in ticks

@@ -236,3 +236,84 @@ def test_stop_kills_paused_container(tmp_path, run):
         await rt.close()
 
     run(main())
+
+
+def test_daemon_restart_adopts_running_container(tmp_path, run):
+    """dockerd containers survive the daemon; the proc runtime must too:
+    a new runtime instance on the same data dir adopts the still-running
+    process (pid + /proc starttime identity) and can stop it."""
+    import os
+
+    from gpu_docker_api_amd.models.etcd import ContainerSpec
+    from gpu_docker_api_amd.runtime.proc import ProcRuntime
+
+    async def main():
+        rt1 = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        spec = ContainerSpec()
+        spec.config = {"Cmd": ["sleep", "60"]}
+        spec.container_name = "adopt-1"
+        await rt1.create(spec)
+        await rt1.start("adopt-1")
+        pid = (await rt1.inspect("adopt-1")).pid
+        assert pid > 0
+        # simulate daemon death: drop the runtime WITHOUT stopping anything
+        if rt1._supervisor is not None:
+            rt1._supervisor.cancel()
+
+        rt2 = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        st = await rt2.inspect("adopt-1")
+        assert st is not None and st.running and st.pid == pid
+        assert os.path.exists(f"/proc/{pid}")
+        stats = await rt2.stats("adopt-1")
+        assert stats["running"] and stats["memoryBytes"] > 0
+        # the adopted process is stoppable through the new runtime
+        await rt2.stop("adopt-1", timeout=5)
+        assert not os.path.exists(f"/proc/{pid}") or open(f"/proc/{pid}/stat").read().split()[2] == "Z"
+        st = await rt2.inspect("adopt-1")
+        assert not st.running
+        await rt2.remove("adopt-1", force=True)
+        await rt2.close()
+
+    run(main())
+
+
+def test_daemon_restart_resurrects_dead_unless_stopped(tmp_path, run):
+    """A container that died while the daemon was down is restarted by the
+    new daemon's supervisor when its policy says unless-stopped (docker's
+    restart-on-daemon-start semantics)."""
+    import asyncio
+    import os
+    import signal as _signal
+
+    from gpu_docker_api_amd.models.etcd import ContainerSpec
+    from gpu_docker_api_amd.runtime.proc import ProcRuntime
+
+    async def main():
+        rt1 = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        spec = ContainerSpec()
+        spec.config = {"Cmd": ["sleep", "60"]}
+        spec.host_config = {"RestartPolicy": {"Name": "unless-stopped"}}
+        spec.container_name = "res-1"
+        await rt1.create(spec)
+        await rt1.start("res-1")
+        pid = (await rt1.inspect("res-1")).pid
+        if rt1._supervisor is not None:
+            rt1._supervisor.cancel()
+        # kill the workload while "no daemon" is watching (it stays a
+        # zombie of THIS process until reaped — the runtime must treat
+        # state Z as dead)
+        os.killpg(pid, _signal.SIGKILL)
+        await asyncio.sleep(0.1)
+
+        rt2 = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        # the new supervisor must resurrect it
+        deadline = asyncio.get_event_loop().time() + 5
+        st = await rt2.inspect("res-1")
+        while asyncio.get_event_loop().time() < deadline and not st.running:
+            await asyncio.sleep(0.1)
+            st = await rt2.inspect("res-1")
+        assert st.running and st.pid != pid
+        await rt2.remove("res-1", force=True)
+        await rt2.close()
+
+    run(main())
