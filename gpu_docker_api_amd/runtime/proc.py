@@ -64,6 +64,7 @@ class _Proc:
         self.attached_pid = 0
         self.attached_start = 0
         self.was_started = False
+        self.start_ticks = 0  # cached /proc starttime of the live leader
 
 
 class ProcRuntime(RuntimeDriver):
@@ -220,9 +221,7 @@ class ProcRuntime(RuntimeDriver):
             meta = {
                 "id": p.state.id,
                 "pid": self._leader_pid(p) if p.state.running else 0,
-                "starttime": (
-                    self._proc_starttime(self._leader_pid(p)) if p.state.running else 0
-                ),
+                "starttime": p.start_ticks if p.state.running else 0,
                 "running": p.state.running,
                 "restarts": p.restarts,
             }
@@ -268,6 +267,7 @@ class ProcRuntime(RuntimeDriver):
                     # the workload outlived the daemon: adopt it live
                     p.attached_pid = pid
                     p.attached_start = int(meta["starttime"])
+                    p.start_ticks = p.attached_start
                     p.was_started = True
                     st.pid = pid
                     st.running, st.status = True, "running"
@@ -325,7 +325,8 @@ class ProcRuntime(RuntimeDriver):
         self._procs[name] = p
         with open(os.path.join(self._cdir(name), "spec.json"), "w") as f:
             f.write(spec.serialize())
-        self._write_meta(p)
+        # meta (pid identity) is written at start(); a created-not-started
+        # container needs none
         return cid
 
     def _get(self, name: str) -> _Proc:
@@ -415,6 +416,7 @@ class ProcRuntime(RuntimeDriver):
             logf.close()
         p.attached_pid = 0
         p.was_started = True
+        p.start_ticks = self._proc_starttime(p.popen.pid) or 0
         p.state.pid = p.popen.pid
         p.state.running, p.state.paused, p.state.status = True, False, "running"
         self._setup_cgroup(name, p, p.popen.pid)
@@ -428,7 +430,7 @@ class ProcRuntime(RuntimeDriver):
         except ProcessLookupError:
             pass
 
-    async def stop(self, name: str, timeout: int = 10) -> None:
+    async def stop(self, name: str, timeout: int = 10, _removing: bool = False) -> None:
         p = self._get(name)
         p.manually_stopped = True
         self._refresh(p)
@@ -470,7 +472,8 @@ class ProcRuntime(RuntimeDriver):
         p.state.running, p.state.paused, p.state.status = False, False, "exited"
         p.state.pid = 0
         p.attached_pid = 0
-        self._write_meta(p)
+        if not _removing:  # the container dir is deleted next anyway
+            self._write_meta(p)
 
     async def pause(self, name: str) -> None:
         p = self._get(name)
@@ -508,7 +511,7 @@ class ProcRuntime(RuntimeDriver):
         if p.state.running:
             if not force:
                 raise RuntimeError(f"{name} is running")
-            await self.stop(name, timeout=2)
+            await self.stop(name, timeout=2, _removing=True)
         if p.cgroup:
             try:
                 os.rmdir(p.cgroup)
