@@ -317,3 +317,46 @@ def test_daemon_restart_resurrects_dead_unless_stopped(tmp_path, run):
         await rt2.close()
 
     run(main())
+
+
+def test_full_daemon_restart_with_live_workload(tmp_path, run):
+    """Daemon-level: a proc-runtime replicaSet keeps RUNNING through a
+    daemon crash; the next daemon adopts it (runtime metadata) AND still
+    knows its versions/allocations (persisted store) — then can exec into
+    it, patch it, and delete it."""
+    import os
+
+    from gpu_docker_api_amd.models import ContainerExecute, ContainerRun, GpuPatch, PatchRequest
+    from gpu_docker_api_amd.routers.app import Daemon
+    from helpers import make_config
+
+    async def main():
+        cfg = make_config(tmp_path, runtime="proc")
+        d1 = Daemon(cfg)
+        await d1.start()
+        await d1.replicaset.run_gpu_container(
+            ContainerRun(image_name="img", replica_set_name="live", gpu_count=1,
+                         cmd=["sleep", "120"])
+        )
+        pid = (await d1.runtime.inspect("live-1")).pid
+        await d1.queue.close()  # flush write-behind state; no graceful stop
+
+        d2 = Daemon(make_config(tmp_path, runtime="proc"), store=d1.store)
+        await d2.start()
+        st = await d2.runtime.inspect("live-1")
+        assert st is not None and st.running and st.pid == pid  # adopted
+        assert sum(d2.gpu.get_gpu_status().values()) == 1       # still allocated
+        out, rc = await d2.replicaset.execute_container(
+            "live", ContainerExecute(cmd=["sh", "-c", "echo adopted-$PPID"])
+        )
+        assert rc == 0 and "adopted-" in out
+        res = await d2.replicaset.patch_container(
+            "live", PatchRequest(gpu_patch=GpuPatch(gpu_count=0))
+        )
+        assert res["containerName"] == "live-2"
+        assert not os.path.exists(f"/proc/{pid}") or \
+            open(f"/proc/{pid}/stat").read().split()[2] == "Z"  # old one stopped
+        await d2.replicaset.delete_container("live")
+        await d2.stop()
+
+    run(main())
