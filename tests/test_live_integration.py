@@ -436,3 +436,67 @@ def test_docker_logs_over_socket(dockerd_sim, tmp_path, run):
         await rt.close()
 
     run(main())
+
+
+def test_engine_restart_preserves_containers(tmp_path, run):
+    """Engine-side durability over the wire: kill the dockerd-sim process
+    and start a new one on the same data dir — the driver must still see
+    the container (adopted live by the new engine), like a real dockerd
+    restart."""
+    from gpu_docker_api_amd.models.etcd import ContainerSpec
+    from gpu_docker_api_amd.runtime.docker import DockerRuntime
+
+    sock1 = str(tmp_path / "d1.sock")
+    data = str(tmp_path / "engine")
+
+    def spawn(sock):
+        return subprocess.Popen(
+            [sys.executable, "-m", "gpu_docker_api_amd.testing.dockerd_sim",
+             "--socket", sock, "--data", data],
+            cwd=REPO,
+            stdout=subprocess.DEVNULL,
+            stderr=subprocess.DEVNULL,
+            start_new_session=True,
+        )
+
+    proc1 = spawn(sock1)
+    proc2 = None
+    try:
+        _wait_for(lambda: os.path.exists(sock1), what="engine socket")
+
+        async def phase1():
+            rt = DockerRuntime(socket_path=sock1)
+            spec = ContainerSpec()
+            spec.container_name = "sv-1"
+            spec.config = {"Image": "img", "Cmd": ["sleep", "60"]}
+            await rt.create(spec)
+            await rt.start("sv-1")
+            st = await rt.inspect("sv-1")
+            await rt.close()
+            return st.pid
+
+        pid = run(phase1())
+        assert pid > 0
+        proc1.kill()
+        proc1.wait(timeout=10)
+
+        sock2 = str(tmp_path / "d2.sock")
+        proc2 = spawn(sock2)
+        _wait_for(lambda: os.path.exists(sock2), what="restarted engine socket")
+
+        async def phase2():
+            rt = DockerRuntime(socket_path=sock2)
+            st = await rt.inspect("sv-1")
+            assert st is not None and st.running and st.pid == pid
+            await rt.remove("sv-1", force=True)
+            await rt.close()
+
+        run(phase2())
+    finally:
+        for p in (proc1, proc2):
+            if p is not None:
+                try:
+                    p.kill()
+                    p.wait(timeout=5)
+                except Exception:
+                    pass
