@@ -360,3 +360,23 @@ def test_full_daemon_restart_with_live_workload(tmp_path, run):
         await d2.stop()
 
     run(main())
+
+
+def test_crash_leftover_deleting_dirs_swept(tmp_path, run):
+    """A crash between the delete-rename and the batched rm leaves a
+    '<name>.deleting-*' dir; the next runtime sweeps it at load."""
+    import os
+
+    from gpu_docker_api_amd.runtime.proc import ProcRuntime
+
+    leftover = tmp_path / "containers" / "old-1.deleting-deadbeef"
+    leftover.mkdir(parents=True)
+    (leftover / "junk.bin").write_bytes(b"x" * 128)
+
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        assert not os.path.exists(leftover)
+        assert await rt.inspect("old-1.deleting-deadbeef") is None
+        await rt.close()
+
+    run(main())
