@@ -369,14 +369,16 @@ def test_crash_leftover_deleting_dirs_swept(tmp_path, run):
 
     from gpu_docker_api_amd.runtime.proc import ProcRuntime
 
-    leftover = tmp_path / "containers" / "old-1.deleting-deadbeef"
+    leftover = tmp_path / "trash" / "old-1-deadbeef"
     leftover.mkdir(parents=True)
     (leftover / "junk.bin").write_bytes(b"x" * 128)
+    legacy = tmp_path / "containers" / "old-2.deleting-cafe"
+    legacy.mkdir(parents=True)
 
     async def main():
         rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
         assert not os.path.exists(leftover)
-        assert await rt.inspect("old-1.deleting-deadbeef") is None
+        assert not os.path.exists(legacy)
         await rt.close()
 
     run(main())
