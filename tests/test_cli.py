@@ -69,6 +69,13 @@ def test_cli_full_lifecycle(live_daemon):
     rows = json.loads(out.stdout)
     assert rows[0]["name"] == "job" and rows[0]["gpuCount"] == 1
 
+    out = _cli(port, "stats", "job")
+    st = json.loads(out.stdout)
+    assert st["running"] is True and len(st["gpus"]) == 1
+
+    out = _cli(port, "logs", "job", "--tail", "5")
+    assert out.returncode == 0  # mock runtime: empty log, command succeeds
+
     out = _cli(port, "patch", "job", "--gpus", "2")
     assert json.loads(out.stdout)["containerName"] == "job-2"
 

@@ -500,3 +500,30 @@ def test_engine_restart_preserves_containers(tmp_path, run):
                     p.wait(timeout=5)
                 except Exception:
                     pass
+
+
+def test_concurrent_clients_over_socket(dockerd_sim, tmp_path, run):
+    """Multiple driver clients hammer the engine concurrently: no
+    cross-talk, all lifecycles complete."""
+    from gpu_docker_api_amd.models.etcd import ContainerSpec
+    from gpu_docker_api_amd.runtime.docker import DockerRuntime
+
+    async def one(i: int):
+        rt = DockerRuntime(socket_path=dockerd_sim)
+        name = f"cc{i}-1"
+        spec = ContainerSpec()
+        spec.container_name = name
+        spec.config = {"Image": "img", "Cmd": ["sleep", "30"]}
+        await rt.create(spec)
+        await rt.start(name)
+        out, rc = await rt.execute_rc(name, ["sh", "-c", f"echo tenant-{i}"])
+        assert rc == 0 and f"tenant-{i}" in out
+        st = await rt.inspect(name)
+        assert st.running and st.name == name
+        await rt.remove(name, force=True)
+        await rt.close()
+
+    async def main():
+        await asyncio.gather(*(one(i) for i in range(6)))
+
+    run(main())
