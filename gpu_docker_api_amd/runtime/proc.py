@@ -59,6 +59,10 @@ class _Proc:
         )
         self.manually_stopped = False
         self.restarts = 0
+        # crash-loop backoff (docker semantics: delay doubles per prompt
+        # death, resets after a stable run)
+        self.next_restart_at = 0.0
+        self.last_start_at = 0.0
         # daemon-restart reattachment: leader pid + its /proc starttime
         # (detects pid reuse) for containers adopted without a Popen handle
         self.attached_pid = 0
@@ -120,6 +124,15 @@ class ProcRuntime(RuntimeDriver):
                     and p.restart_policy in ("always", "unless-stopped")
                     and (p.popen is not None or p.was_started)
                 ):
+                    now = time.monotonic()
+                    if now < p.next_restart_at:
+                        continue
+                    # exponential crash-loop backoff (docker: 100 ms
+                    # doubling, capped; reset after >=10 s of stable run)
+                    if p.last_start_at and now - p.last_start_at >= 10.0:
+                        p.restarts = 0
+                    delay = min(0.1 * (2 ** min(p.restarts, 8)), 30.0)
+                    p.next_restart_at = now + delay
                     try:
                         p.restarts += 1
                         await self.start(name)
@@ -425,6 +438,7 @@ class ProcRuntime(RuntimeDriver):
             logf.close()
         p.attached_pid = 0
         p.was_started = True
+        p.last_start_at = time.monotonic()
         p.start_ticks = self._proc_starttime(p.popen.pid) or 0
         p.state.pid = p.popen.pid
         p.state.running, p.state.paused, p.state.status = True, False, "running"

@@ -382,3 +382,34 @@ def test_crash_leftover_deleting_dirs_swept(tmp_path, run):
         await rt.close()
 
     run(main())
+
+
+def test_crash_loop_backoff(tmp_path, run):
+    """A workload that dies instantly must not be restarted at full
+    supervisor tick rate: the restart counter must grow slower than the
+    tick rate (docker's exponential crash-loop backoff)."""
+    import asyncio
+
+    from gpu_docker_api_amd.models.etcd import ContainerSpec
+    from gpu_docker_api_amd.runtime.proc import ProcRuntime
+
+    async def main():
+        rt = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        spec = ContainerSpec()
+        spec.config = {"Cmd": ["false"]}  # exits immediately, rc 1
+        spec.host_config = {"RestartPolicy": {"Name": "always"}}
+        spec.container_name = "loop-1"
+        await rt.create(spec)
+        await rt.start("loop-1")
+        await asyncio.sleep(2.0)
+        p = rt._procs["loop-1"]
+        # ~10 ticks/2s without backoff would mean ~10 restarts; with
+        # 0.1*2^n backoff the schedule is 0.2+0.4+0.8+... -> at most 4-5
+        assert 1 <= p.restarts <= 6, p.restarts
+        first = p.restarts
+        await asyncio.sleep(1.5)
+        assert p.restarts - first <= 2  # slowing down, not tick-rate
+        await rt.remove("loop-1", force=True)
+        await rt.close()
+
+    run(main())
