@@ -291,12 +291,17 @@ class ProcRuntime(RuntimeDriver):
                     p.attached_start = int(meta["starttime"])
                     p.start_ticks = p.attached_start
                     p.was_started = True
+                    # treat adoption as a fresh stable-run baseline so a
+                    # later death gets normal (not stale) backoff
+                    p.last_start_at = time.monotonic()
                     st.pid = pid
                     st.running, st.status = True, "running"
                 else:
                     # it died while unsupervised; the restart policy decides
                     # whether the supervisor resurrects it (docker restarts
-                    # unless-stopped containers on daemon start)
+                    # unless-stopped containers on daemon start, without
+                    # carrying the previous daemon's crash-loop backoff)
+                    p.restarts = 0
                     p.was_started = True
                     p.manually_stopped = False
             else:
