@@ -632,7 +632,11 @@ class ReplicaSetService:
         out = await self.runtime.stats(self._current_vname(name))
         if spec.gpu_uuids:
             try:
-                fresh = {g.uuid: g for g in (self.gpu.inventory.refresh_usage() or [])}
+                # amdsmi reads can take tens of ms: off the event loop
+                gpus = await asyncio.get_running_loop().run_in_executor(
+                    None, self.gpu.inventory.refresh_usage
+                )
+                fresh = {g.uuid: g for g in (gpus or [])}
             except Exception:  # inventory backend without live usage
                 fresh = {}
             out["gpus"] = [
