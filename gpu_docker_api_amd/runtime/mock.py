@@ -138,7 +138,8 @@ class MockRuntime(RuntimeDriver):
         if name not in self.containers:
             raise ContainerNotExist(name)
         st = self.containers[name]
-        return {"running": st.running, "cpuSeconds": 0.0, "memoryBytes": 0, "pids": 0}
+        return {"running": st.running, "cpuSeconds": 0.0, "memoryBytes": 0,
+                "pids": 0, "restarts": 0}
 
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         self._maybe_fail("commit")

@@ -595,7 +595,7 @@ class ProcRuntime(RuntimeDriver):
         p = self._get(name)
         self._refresh(p)
         out: Dict = {"running": p.state.running, "cpuSeconds": 0.0,
-                     "memoryBytes": 0, "pids": 0}
+                     "memoryBytes": 0, "pids": 0, "restarts": p.restarts}
         if not p.state.running or not self._leader_pid(p):
             return out
         if p.cgroup:
