@@ -8,6 +8,10 @@ state persists under ``cpus/cpuStatusMapKey`` as
 Fixes the reference's allocation bug (cpuscheduler.go:94: iterating slice
 *indices* instead of sorted cpu ids — only accidentally correct for
 contiguous ids): we sort numerically and allocate the lowest free ids.
+
+MI355X extension: ``apply(n, preferred_nodes=[...])`` prefers CPUs on the
+NUMA node(s) of the container's GPUs (sysfs cpu->node map), so
+host<->device staging stays socket-local; falls back to any free CPUs.
 """
 from __future__ import annotations
 
@@ -26,16 +30,31 @@ class CpuScheduler(BaseScheduler):
     resource = Resource.CPUS
     state_key = CPU_STATUS_MAP_KEY
 
-    def __init__(self, store: StateStore, queue: Optional[WorkQueue], count: int = 0) -> None:
+    def __init__(
+        self,
+        store: StateStore,
+        queue: Optional[WorkQueue],
+        count: int = 0,
+        cpu_nodes: Optional[Dict[int, int]] = None,
+    ) -> None:
         super().__init__(store, queue)
         self.available_cpu_nums = count or (os.cpu_count() or 1)
         self.cpu_status_map: Dict[str, int] = {}
+        if cpu_nodes is None:
+            from .numa import cpu_node_map
+
+            cpu_nodes = cpu_node_map()
+        self.cpu_nodes: Dict[int, int] = cpu_nodes or {}
 
     @classmethod
     async def create(
-        cls, store: StateStore, queue: Optional[WorkQueue], count: int = 0
+        cls,
+        store: StateStore,
+        queue: Optional[WorkQueue],
+        count: int = 0,
+        cpu_nodes: Optional[Dict[int, int]] = None,
     ) -> "CpuScheduler":
-        self = cls(store, queue, count)
+        self = cls(store, queue, count, cpu_nodes)
         kv = await store.get_or_none(self.resource, self.state_key)
         if kv is not None:
             data = json.loads(kv.value)
@@ -58,8 +77,10 @@ class CpuScheduler(BaseScheduler):
                 separators=(",", ":"),
             )
 
-    def apply(self, num: int) -> str:
-        """Allocate ``num`` cpus; returns 'i,j,k' for CpusetCpus."""
+    def apply(self, num: int, preferred_nodes: Optional[List[int]] = None) -> str:
+        """Allocate ``num`` cpus; returns 'i,j,k' for CpusetCpus.
+        ``preferred_nodes``: NUMA nodes to draw from first (the allocated
+        GPUs' nodes) — spills to other nodes only when they run dry."""
         if num <= 0 or num > self.available_cpu_nums:
             raise CpuNotEnough(f"requested {num}, node has {self.available_cpu_nums}")
         with self._lock:
@@ -68,7 +89,14 @@ class CpuScheduler(BaseScheduler):
             )
             if len(free) < num:
                 raise CpuNotEnough(f"requested {num}, only {len(free)} free")
-            chosen = free[:num]
+            if preferred_nodes and self.cpu_nodes:
+                pref = set(preferred_nodes)
+                local = [c for c in free if self.cpu_nodes.get(c) in pref]
+                rest = [c for c in free if self.cpu_nodes.get(c) not in pref]
+                chosen = (local + rest)[:num]
+                chosen.sort()
+            else:
+                chosen = free[:num]
             for c in chosen:
                 self.cpu_status_map[str(c)] = 1
             self._persist_async()

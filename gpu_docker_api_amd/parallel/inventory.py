@@ -41,6 +41,7 @@ class GpuInfo:
     name: str = "AMD Instinct MI355X"
     vram_total: int = MI355X_HBM_BYTES
     vram_used: int = 0
+    numa_node: int = -1     # host NUMA node of the GPU (-1 = unknown)
     render_node: str = ""   # /dev/dri/renderD<N>
     card_node: str = ""     # /dev/dri/card<N>
 
@@ -52,6 +53,7 @@ class GpuInfo:
             "name": self.name,
             "vramTotal": self.vram_total,
             "vramUsed": self.vram_used,
+            "numaNode": self.numa_node,
             "renderNode": self.render_node,
             "cardNode": self.card_node,
         }
@@ -71,6 +73,15 @@ class GpuInventory:
         """Symmetric GB/s matrix; diagonal 0. Static topology estimate —
         measured numbers from the HIP probe overlay this (topology.py)."""
         raise NotImplementedError
+
+
+def _bdf_numa_node(bdf: str) -> int:
+    """Host NUMA node of a PCIe device (-1 when sysfs doesn't say)."""
+    try:
+        with open(f"/sys/bus/pci/devices/{bdf}/numa_node") as f:
+            return int(f.read().strip())
+    except (OSError, ValueError):
+        return -1
 
 
 def _bdf_to_drm_nodes(bdf: str) -> tuple[str, str]:
@@ -139,6 +150,7 @@ class AmdSmiInventory(GpuInventory):
             except Exception:
                 pass
             render, card = _bdf_to_drm_nodes(bdf) if bdf else ("", "")
+            numa = _bdf_numa_node(bdf) if bdf else -1
             out.append(
                 GpuInfo(
                     index=i,
@@ -147,6 +159,7 @@ class AmdSmiInventory(GpuInventory):
                     name=name or "AMD Instinct MI355X",
                     vram_total=total,
                     vram_used=used,
+                    numa_node=numa,
                     render_node=render,
                     card_node=card,
                 )
@@ -197,6 +210,7 @@ class AmdSmiInventory(GpuInventory):
             idx = int(item.get("gpu", i))
             bdf = str(item.get("bdf", ""))
             render, card = _bdf_to_drm_nodes(bdf) if bdf else ("", "")
+            numa = _bdf_numa_node(bdf) if bdf else -1
             name = "AMD Instinct MI355X"
             total = MI355X_HBM_BYTES
             st = static_by_gpu.get(idx) or {}
@@ -215,6 +229,7 @@ class AmdSmiInventory(GpuInventory):
                     bdf=bdf,
                     name=name,
                     vram_total=total,
+                    numa_node=numa,
                     render_node=render,
                     card_node=card,
                 )
@@ -269,6 +284,9 @@ class MockInventory(GpuInventory):
                 uuid=f"MockMI355X-{i}",
                 bdf=f"0000:{0x10 + i:02x}:00.0",
                 vram_used=self._used.get(i, 0),
+                # synthetic NUMA topology (two sockets, 4 GPUs each) so
+                # NUMA-aware cpuset placement is testable on CPU boxes
+                numa_node=i // 4,
                 # synthetic DRM nodes (renderD128+i mirrors amdgpu's
                 # numbering) so the docker device-injection path is
                 # exercised by the mock backend too

@@ -163,7 +163,9 @@ class ReplicaSetService:
             else:
                 spec.gpu_uuids = []
             if req.cpu_count > 0:
-                cpuset = self.cpu.apply(req.cpu_count)
+                cpuset = self.cpu.apply(
+                    req.cpu_count, preferred_nodes=self._gpu_numa_nodes(spec.gpu_uuids)
+                )
                 saga.push("cpu", lambda: self.cpu.restore(cpuset))
                 spec.cpuset_cpus = cpuset
             if req.memory:
@@ -173,6 +175,15 @@ class ReplicaSetService:
             saga.commit()
         phases = timer.finish()
         return {"name": vname, "id": cid, "phases": phases}
+
+    def _gpu_numa_nodes(self, uuids: List[str]) -> List[int]:
+        """NUMA nodes of the allocated GPUs (for socket-local cpusets)."""
+        nodes = []
+        for u in uuids or []:
+            info = self.gpu.info_by_uuid(u)
+            if info is not None and info.numa_node >= 0:
+                nodes.append(info.numa_node)
+        return sorted(set(nodes))
 
     async def _exist_container_any(self, name: str) -> bool:
         """True if any versioned container of this replicaSet exists (the
@@ -313,7 +324,11 @@ class ReplicaSetService:
         if old_set:
             self.cpu.restore(old_set)
             saga.push("cpu-reacquire-old", lambda: self.cpu.apply_specific(old_set))
-        new_cpuset = self.cpu.apply(new_count) if new_count > 0 else ""
+        new_cpuset = (
+            self.cpu.apply(new_count, preferred_nodes=self._gpu_numa_nodes(spec.gpu_uuids))
+            if new_count > 0
+            else ""
+        )
         if new_cpuset:
             saga.push("cpu-release-new", lambda: self.cpu.restore(new_cpuset))
         spec.cpuset_cpus = new_cpuset
@@ -536,7 +551,10 @@ class ReplicaSetService:
                     saga.push("gpu", lambda: self.gpu.restore(uuids))
                     spec.gpu_uuids = uuids
                 if spec.cpuset_cpus:
-                    cpuset = self.cpu.apply(len(spec.cpuset_cpus.split(",")))
+                    cpuset = self.cpu.apply(
+                        len(spec.cpuset_cpus.split(",")),
+                        preferred_nodes=self._gpu_numa_nodes(spec.gpu_uuids),
+                    )
                     saga.push("cpu", lambda: self.cpu.restore(cpuset))
                     spec.cpuset_cpus = cpuset
                 self._released.discard(vname_old)

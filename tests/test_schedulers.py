@@ -230,3 +230,38 @@ def test_probe_cache_file_roundtrip(tmp_path):
     # corrupt file degrades to None, not a crash
     p.write_text("{not json")
     assert Topology.load_probe_file(str(p)) is None
+
+
+def test_cpu_numa_preferred_allocation(run):
+    """MI355X extension: cpuset drawn from the GPUs' NUMA node first,
+    spilling to other nodes only when the preferred ones run dry."""
+    from gpu_docker_api_amd.parallel.cpu import CpuScheduler
+    from gpu_docker_api_amd.state.store import MemoryStore
+
+    async def main():
+        store = MemoryStore()
+        nodes = {c: (0 if c < 4 else 1) for c in range(8)}
+        cpu = await CpuScheduler.create(store, None, count=8, cpu_nodes=nodes)
+        assert cpu.apply(2, preferred_nodes=[1]) == "4,5"
+        # spill: node 1 has only 2 free left -> take them + lowest others
+        assert cpu.apply(4, preferred_nodes=[1]) == "0,1,6,7"
+        cpu.restore(["0", "1", "4", "5", "6", "7"])
+        # no preference -> lowest ids
+        assert cpu.apply(3) == "0,1,2"
+
+    run(main())
+
+
+def test_gpu_numa_nodes_resolution(tmp_path, run):
+    from gpu_docker_api_amd.models import ContainerRun
+    from helpers import make_daemon
+
+    async def main():
+        d = await make_daemon(tmp_path)
+        # mock inventory: GPUs 0-3 node 0, GPUs 4-7 node 1
+        assert d.replicaset._gpu_numa_nodes(["MockMI355X-1", "MockMI355X-6"]) == [0, 1]
+        assert d.replicaset._gpu_numa_nodes(["MockMI355X-2"]) == [0]
+        assert d.replicaset._gpu_numa_nodes([]) == []
+        await d.stop()
+
+    run(main())
