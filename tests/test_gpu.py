@@ -399,3 +399,36 @@ def test_tenant_gpu_workload_end_to_end(tmp_path, run):
     import asyncio
 
     run(main())
+
+
+def test_numa_local_cpuset_on_hardware(tmp_path, run):
+    """A 1-GPU container's cpuset must come from the GPU's NUMA node
+    (sysfs truth) when the host exposes multiple nodes."""
+    require_gpu()
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from helpers import make_daemon
+
+    from gpu_docker_api_amd.models import ContainerRun
+    from gpu_docker_api_amd.parallel.numa import cpu_node_map
+
+    async def main():
+        node_of = cpu_node_map()
+        d = await make_daemon(tmp_path, runtime="proc", inventory="amdsmi")
+        gpu_nodes = {g.numa_node for g in d.gpu.gpus if g.numa_node >= 0}
+        if len(set(node_of.values())) < 2 or not gpu_nodes:
+            pytest.skip("host exposes no multi-node NUMA topology")
+        await d.replicaset.run_gpu_container(
+            ContainerRun(image_name="img", replica_set_name="nm", gpu_count=1,
+                         cpu_count=4)
+        )
+        st = await d.runtime.inspect("nm-1")
+        cpus = [int(c) for c in st.cpuset_cpus.split(",")]
+        alloc_nodes = {node_of.get(c) for c in cpus}
+        spec_gpu_nodes = d.replicaset._gpu_numa_nodes(st.gpu_uuids)
+        assert alloc_nodes == set(spec_gpu_nodes), (cpus, alloc_nodes, spec_gpu_nodes)
+        await d.replicaset.delete_container("nm")
+        await d.stop()
+
+    run(main())
