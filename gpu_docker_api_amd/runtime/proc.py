@@ -410,6 +410,13 @@ class ProcRuntime(RuntimeDriver):
             if p.state.cpuset_cpus:
                 with open(os.path.join(cg, "cpuset.cpus"), "w") as f:
                     f.write(p.state.cpuset_cpus)
+            mems = p.spec.host_config.get("CpusetMems", "")
+            if mems:
+                try:
+                    with open(os.path.join(cg, "cpuset.mems"), "w") as f:
+                        f.write(mems)
+                except OSError:
+                    pass  # node went away / not delegated: cpus still bound
             with open(os.path.join(cg, "cgroup.procs"), "w") as f:
                 f.write(str(pid))
             p.cgroup = cg

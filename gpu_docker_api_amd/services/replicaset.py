@@ -163,11 +163,14 @@ class ReplicaSetService:
             else:
                 spec.gpu_uuids = []
             if req.cpu_count > 0:
-                cpuset = self.cpu.apply(
-                    req.cpu_count, preferred_nodes=self._gpu_numa_nodes(spec.gpu_uuids)
-                )
+                nodes = self._gpu_numa_nodes(spec.gpu_uuids)
+                cpuset = self.cpu.apply(req.cpu_count, preferred_nodes=nodes)
                 saga.push("cpu", lambda: self.cpu.restore(cpuset))
                 spec.cpuset_cpus = cpuset
+                if nodes:
+                    # memory locality too (docker HostConfig.CpusetMems; the
+                    # proc runtime writes cgroup cpuset.mems)
+                    spec.host_config["CpusetMems"] = ",".join(map(str, nodes))
             if req.memory:
                 spec.memory_bytes = to_bytes(req.memory)
             timer.mark("schedule")

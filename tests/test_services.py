@@ -516,3 +516,23 @@ def test_container_stats_includes_gpu_hbm(tmp_path, run):
         await d.stop()
 
     run(main())
+
+
+def test_cpuset_mems_recorded_with_gpu_nodes(tmp_path, run):
+    from gpu_docker_api_amd.models import ContainerRun
+
+    async def main():
+        from helpers import make_daemon
+
+        d = await make_daemon(tmp_path)
+        await d.replicaset.run_gpu_container(
+            ContainerRun(image_name="img", replica_set_name="mm", gpu_count=2,
+                         cpu_count=2)
+        )
+        spec = await d.replicaset._load_spec("mm")
+        nodes = d.replicaset._gpu_numa_nodes(spec.gpu_uuids)
+        assert spec.host_config.get("CpusetMems") == ",".join(map(str, nodes))
+        await d.replicaset.delete_container("mm")
+        await d.stop()
+
+    run(main())
