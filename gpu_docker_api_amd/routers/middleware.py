@@ -65,7 +65,10 @@ class AuthMiddleware:
             if k == b"authorization":
                 token = v[7:] if v[:7].lower() == b"bearer " else v
                 break
-        if token != self.apikey:
+        import hmac
+
+        # constant-time compare: a == on secrets leaks length/prefix timing
+        if not hmac.compare_digest(token, self.apikey):
             body = json.dumps(
                 {"code": 403, "msg": "Forbidden", "data": None}
             ).encode()
