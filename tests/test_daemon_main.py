@@ -140,3 +140,66 @@ def test_daemon_against_live_etcd_gateway(tmp_path, run):
     run(main())
     server.should_exit = True
     t.join(timeout=10)
+
+
+def test_daemon_sigterm_persists_state_across_processes(tmp_path):
+    """SIGTERM -> uvicorn lifespan shutdown -> Daemon.stop persists; a new
+    daemon PROCESS on the same dataDir recovers allocations, versions and
+    specs from the WAL (the reference's etcd-reload path, main.go:139-154,
+    without the external etcd)."""
+
+    def spawn(port):
+        return subprocess.Popen(
+            [sys.executable, "-m", "gpu_docker_api_amd",
+             "--addr", f"127.0.0.1:{port}",
+             "--runtime", "mock", "--inventory", "mock",
+             "--dataDir", str(tmp_path / "data"),
+             "--portRange", "49300-49400",
+             "--logLevel", "warning"],
+            cwd=ROOT,
+            stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT,
+            start_new_session=True,
+        )
+
+    port1 = _free_port()
+    p1 = spawn(port1)
+    p2 = None
+    try:
+        assert _wait_ping(port1), "daemon 1 did not come up"
+        r = httpx.post(
+            f"http://127.0.0.1:{port1}/api/v1/replicaSet",
+            json={"imageName": "img", "replicaSetName": "dur", "gpuCount": 3},
+            timeout=30,
+        ).json()
+        assert r["code"] == 200
+        p1.terminate()  # SIGTERM: graceful lifespan shutdown
+        assert p1.wait(timeout=20) is not None
+
+        port2 = _free_port()
+        p2 = spawn(port2)
+        assert _wait_ping(port2), "daemon 2 did not come up"
+        gpus = httpx.get(
+            f"http://127.0.0.1:{port2}/api/v1/resources/gpus", timeout=10
+        ).json()
+        assert sum(gpus["data"].values()) == 3  # allocation survived
+        info = httpx.get(
+            f"http://127.0.0.1:{port2}/api/v1/replicaSet/dur", timeout=10
+        ).json()
+        assert info["code"] == 200 and info["data"]["version"] == 1
+        hist = httpx.get(
+            f"http://127.0.0.1:{port2}/api/v1/replicaSet/dur/history", timeout=10
+        ).json()
+        assert [h["version"] for h in hist["data"]] == [1]
+    finally:
+        for p in (p1, p2):
+            if p is not None:
+                try:
+                    p.terminate()
+                    p.wait(timeout=10)
+                except Exception:
+                    try:
+                        p.kill()
+                        p.wait(timeout=5)
+                    except Exception:
+                        pass
