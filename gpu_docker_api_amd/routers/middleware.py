@@ -7,6 +7,8 @@ latter re-buffers every request/response through an ASGI sub-app and costs
 """
 from __future__ import annotations
 
+import hmac
+
 import json
 
 _CORS_HEADERS = [
@@ -65,8 +67,6 @@ class AuthMiddleware:
             if k == b"authorization":
                 token = v[7:] if v[:7].lower() == b"bearer " else v
                 break
-        import hmac
-
         # constant-time compare: a == on secrets leaks length/prefix timing
         if not hmac.compare_digest(token, self.apikey):
             body = json.dumps(
