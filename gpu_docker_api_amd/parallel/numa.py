@@ -10,7 +10,7 @@ from __future__ import annotations
 import glob
 import os
 import re
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 
 def _parse_cpulist(text: str) -> List[int]:
