@@ -7,7 +7,7 @@ point-to-point by xGMI (7 links per GPU, ≈153 GB/s each) and ring collectives
 are per-link bound, so a multi-GPU allocation should maximize intra-set link
 bandwidth. This module holds the adjacency matrix — a static estimate from
 amdsmi link types, overlaid by *measured* numbers from the native HIP
-bandwidth probe (csrc/xgmi_probe.hip via ops.hipcore) when available — and
+bandwidth probe (csrc/hipcore.hip p2p_matrix via ops.hipcore) when available — and
 scores candidate GPU subsets for the scheduler's bin-pack.
 """
 from __future__ import annotations
