@@ -281,6 +281,43 @@ def _mount(app: FastAPI, d: Daemon) -> None:
 
         return StreamingResponse(gen(), media_type="text/event-stream")
 
+    @app.get("/api/v1/images")
+    async def images_list():
+        """MI355X extension: the runtime's image store (proc/mock: seed
+        dirs + commit snapshots; docker: the engine's image list)."""
+        try:
+            data = await d.runtime.image_list()
+        except Exception as exc:  # noqa: BLE001
+            return error(Code.SERVER_BUSY, detail=str(exc))
+        return success(data)
+
+    @app.put("/api/v1/images/{ref}")
+    async def images_import(ref: str, request: Request):
+        """MI355X extension: register a local directory as an image — the
+        'pull it locally first' step the reference delegates to dockerd
+        (its OpenAPI: imageName 'is not automatically downloaded')."""
+        import re as _re
+
+        # image refs: name[:tag] with registry-ish chars; traversal-safe —
+        # the runtime's _image_dir additionally realpath-contains the
+        # flattened ref
+        if not _re.fullmatch(r"[A-Za-z0-9_][A-Za-z0-9_.:/\-]{0,127}", ref) or ".." in ref:
+            return error(Code.INVALID_PARAMS, detail=f"invalid image ref {ref!r}")
+        try:
+            body = await request.json()
+        except Exception:
+            body = {}
+        path = (body or {}).get("path", "")
+        if not path:
+            return error(Code.INVALID_PARAMS, detail="body needs {path: <local dir>}")
+        try:
+            out = await d.runtime.image_import(ref, path)
+        except FileNotFoundError:
+            return error(Code.INVALID_PARAMS, detail=f"not a directory: {path}")
+        except Exception as exc:  # noqa: BLE001
+            return error(Code.SERVER_BUSY, detail=str(exc))
+        return success({"ref": out})
+
     @app.post("/api/v1/admin/compact")
     async def compact(request: Request):
         """MI355X extension: discard state history below a revision and

@@ -113,6 +113,15 @@ class RuntimeDriver:
         the reference has no stats route)."""
         raise NotImplementedError
 
+    async def image_import(self, ref: str, src_path: str) -> str:
+        """Register a local directory as image ``ref`` (proc/mock: the
+        'pull it locally first' step the reference delegates to dockerd).
+        Drivers whose engine owns images raise RuntimeError."""
+        raise NotImplementedError
+
+    async def image_list(self) -> List[Dict]:
+        raise NotImplementedError
+
     # ---- volumes ----
     async def volume_create(
         self, name: str, driver_opts: Optional[Dict[str, str]] = None

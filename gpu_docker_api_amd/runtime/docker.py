@@ -238,6 +238,18 @@ class DockerRuntime(RuntimeDriver):
         )
         return self._demux_stream(raw or b"")
 
+    async def image_import(self, ref: str, src_path: str) -> str:
+        raise RuntimeError(
+            "the docker engine owns its image store: docker pull/load there"
+        )
+
+    async def image_list(self) -> List[Dict[str, Any]]:
+        items = await self._call("GET", "/images/json") or []
+        return [
+            {"ref": (it.get("RepoTags") or [it.get("Id", "")])[0], "id": it.get("Id", "")}
+            for it in items
+        ]
+
     async def stats(self, name: str) -> Dict[str, Any]:
         d = await self._call(
             "GET", f"/containers/{name}/stats",

@@ -141,6 +141,17 @@ class MockRuntime(RuntimeDriver):
         return {"running": st.running, "cpuSeconds": 0.0, "memoryBytes": 0,
                 "pids": 0, "restarts": 0}
 
+    async def image_import(self, ref: str, src_path: str) -> str:
+        import os as _os
+
+        if not _os.path.isdir(src_path):
+            raise FileNotFoundError(src_path)
+        self.images[ref] = src_path
+        return ref
+
+    async def image_list(self) -> list:
+        return [{"ref": r, "path": p} for r, p in sorted(self.images.items())]
+
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         self._maybe_fail("commit")
         st = self._get(name)

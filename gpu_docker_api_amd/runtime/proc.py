@@ -650,6 +650,27 @@ class ProcRuntime(RuntimeDriver):
         lines = data.decode(errors="replace").splitlines()
         return "\n".join(lines[-tail:]) + ("\n" if lines else "")
 
+    async def image_import(self, ref: str, src_path: str) -> str:
+        if not os.path.isdir(src_path):
+            raise FileNotFoundError(src_path)
+        dest = self._image_dir(ref)
+
+        def _snapshot():
+            shutil.rmtree(dest, ignore_errors=True)
+            shutil.copytree(src_path, dest, symlinks=True)
+
+        await asyncio.get_running_loop().run_in_executor(None, _snapshot)
+        return ref
+
+    async def image_list(self) -> List[Dict]:
+        iroot = os.path.join(self.base, "images")
+        out = []
+        for entry in sorted(os.listdir(iroot)):
+            path = os.path.join(iroot, entry)
+            if os.path.isdir(path):
+                out.append({"ref": entry, "path": path})
+        return out
+
     async def commit(self, name: str, image: str, tag: str = "") -> str:
         p = self._get(name)
         ref = f"{image}:{tag}" if tag else image

@@ -69,6 +69,8 @@ OUR_EXTENSION_ROUTES = {
     ("POST", "/api/v1/resources/gpus/validate"): "MFMA burn-in validation of a GPU set before handing it to a tenant",
     ("GET", "/api/v1/events"): "live state-change stream (SSE)",
     ("POST", "/api/v1/admin/compact"): "explicit history compaction (etcd compaction analog)",
+    ("GET", "/api/v1/images"): "runtime image store listing",
+    ("PUT", "/api/v1/images/{name}"): "register a local dir as an image (the reference delegates 'pull it locally first' to dockerd)",
     ("GET", "/metrics"): "Prometheus latency/phase metrics",
     ("GET", "/ping"): "health probe (reference has it too, outside its OpenAPI: main.go:119-123)",
 }

@@ -204,3 +204,38 @@ def test_metrics_endpoint(client):
     text = client.get("/metrics").text
     assert "gda_replicaset_run_count" in text
     assert "gda_replicaset_run_p50_ms" in text
+
+
+def test_image_import_and_run_from_seed(tmp_path, run):
+    """Extension: PUT /images/{ref} registers a local dir; a replicaSet
+    created from that image materializes its rootfs from the seed."""
+    from fastapi.testclient import TestClient
+
+    from gpu_docker_api_amd.routers.app import build_app
+    from helpers import make_config
+
+    seed = tmp_path / "seed"
+    seed.mkdir()
+    (seed / "hello.txt").write_text("from-the-image")
+
+    app = build_app(make_config(tmp_path, runtime="proc"))
+    with TestClient(app) as client:
+        r = client.put("/api/v1/images/base:v1", json={"path": str(seed)}).json()
+        assert r["code"] == 200 and r["data"]["ref"] == "base:v1"
+        r = client.get("/api/v1/images").json()
+        assert any(i["ref"].startswith("base") for i in r["data"])
+        r = client.post(
+            "/api/v1/replicaSet",
+            json={"imageName": "base:v1", "replicaSetName": "seeded", "gpuCount": 0,
+                  "cmd": ["sleep", "30"]},
+        ).json()
+        assert r["code"] == 200
+        r = client.post(
+            "/api/v1/replicaSet/seeded/execute",
+            json={"cmd": ["cat", "hello.txt"]},
+        ).json()
+        assert r["code"] == 200 and "from-the-image" in r["data"]["stdout"]
+        # bad path rejected
+        r = client.put("/api/v1/images/x", json={"path": str(tmp_path / "nope")}).json()
+        assert r["code"] != 200
+        client.delete("/api/v1/replicaSet/seeded")
