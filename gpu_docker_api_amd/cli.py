@@ -204,6 +204,23 @@ def vol_delete(name: str, keep_record: bool = typer.Option(False, "--keep-record
         _show(c.delete(f"/api/v1/volumes/{name}", params=params))
 
 
+img = typer.Typer(help="runtime image store")
+app.add_typer(img, name="image")
+
+
+@img.command("ls")
+def image_ls():
+    with _client() as c:
+        _show(c.get("/api/v1/images"))
+
+
+@img.command("import")
+def image_import(ref: str, path: str):
+    """Register a local directory as an image (proc/mock runtimes)."""
+    with _client() as c:
+        _show(c.put(f"/api/v1/images/{ref}", json={"path": path}))
+
+
 res = typer.Typer(help="node resources")
 app.add_typer(res, name="resources")
 
