@@ -31,6 +31,17 @@
 // operand; cbsz=blgp=4) was verified on hardware with the same
 // single-MFMA exact-integer probe as fp8 (max abs err 0.0).
 //
+// Shipped defaults (each measured in both A/B interleave orders; the
+// tuning ladder with PMC evidence is profiles/gemm_kernel_stats.md):
+//   fp8: SWZV=1 (row-bit-3 swizzle; bank conflicts 44M -> 0, +11%),
+//        unmerged phases — 2.22 PF @8192^3 random operands;
+//   fp4: SWZV=0, MP23=1 (merged phases 2+3, +1-2%) — 4.02-4.09 PF,
+//        bit-exact vs the dequantized fp32 reference.
+// Rejected-by-measurement variants stay selectable for A/B: shape 17
+// (alternate swizzle), 18 (quad-transpose dwordx4 epilogue), 19/20
+// (merge toggles), 32 (the 32x32x64 instruction: only 2 independent
+// accumulators per phase against its 64-cycle dependent latency).
+//
 // Constraints: M,N multiples of 256; K multiple of 128 (fp8) / 256 (fp4),
 // K >= 2 K-steps.
 #include <hip/hip_runtime.h>
