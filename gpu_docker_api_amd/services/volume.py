@@ -15,6 +15,8 @@ with its bugs fixed:
 """
 from __future__ import annotations
 
+import asyncio
+
 import logging
 from typing import Dict, List, Optional
 
@@ -111,7 +113,19 @@ class VolumeService:
                 "DriverOpts": driver_opts,
             },
         )
+        # compensate the spec write too: a failed later step (migration)
+        # must not leave a v+1 spec disagreeing with the reverted version
+        # map (same class of gap ADVICE r1 #2 found on the container side)
+        prev = await self.store.get_or_none(Resource.VOLUMES, name)
+
+        async def _undo_persist(prev=prev):
+            if prev is None:
+                await self.store.delete(Resource.VOLUMES, name)
+            else:
+                await self.store.put(Resource.VOLUMES, name, prev.value)
+
         await self._persist(name, spec)
+        saga.push("spec", _undo_persist)
         return spec, vname
 
     # ----------------------------------------------------------------- resize
@@ -131,7 +145,10 @@ class VolumeService:
             raise NoPatchRequired(f"{name}: same size")
         old_vs = await self.runtime.volume_inspect(vname_old)
         if new_bytes < old_bytes and old_vs is not None:
-            used = dir_size(old_vs.mountpoint)
+            # fs walk of a large volume must not stall the event loop
+            used = await asyncio.get_running_loop().run_in_executor(
+                None, dir_size, old_vs.mountpoint
+            )
             if used > new_bytes:
                 raise VolumeSizeUsedGreaterThanReduced(
                     f"{name}: used {used} > requested {new_bytes}"

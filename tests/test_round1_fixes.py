@@ -223,3 +223,35 @@ def test_valid_name_implies_safe_subpath(tmp_path):
             assert os.path.basename(p) == name
 
     check()
+
+
+def test_failed_volume_resize_restores_spec(tmp_path, run):
+    """Same compensation class as the container side: a resize whose
+    migration fails must leave the VOLUMES spec at the previous version,
+    agreeing with the reverted version map."""
+    from gpu_docker_api_amd.models import VolumeCreate
+
+    async def main():
+        d = await make_daemon(tmp_path)
+        await d.volume.create_volume(VolumeCreate(name="v", size="1GB"))
+        assert d.volume_versions.get("v") == 1
+
+        orig = d.replicaset.copy.move_contents
+
+        async def boom(src, dest):
+            raise RuntimeError("injected migrate failure")
+
+        d.volume.copy.move_contents = boom
+        with pytest.raises(RuntimeError):
+            await d.volume.patch_volume_size("v", "2GB")
+        d.volume.copy.move_contents = orig
+
+        assert d.volume_versions.get("v") == 1  # version map reverted
+        spec = await d.volume._load_spec("v")
+        assert spec.version == 1 and spec.size == "1GB"  # spec agrees
+        # and a later resize still works cleanly
+        out = await d.volume.patch_volume_size("v", "2GB")
+        assert out["name"] == "v-2"
+        await d.stop()
+
+    run(main())
