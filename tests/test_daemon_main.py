@@ -203,3 +203,37 @@ def test_daemon_sigterm_persists_state_across_processes(tmp_path):
                         p.wait(timeout=5)
                     except Exception:
                         pass
+
+
+def test_config_file_layering(tmp_path):
+    """YAML config (the reference documents etc/config.yaml but never
+    implemented it) with flag precedence: defaults < file < flags."""
+    from gpu_docker_api_amd.__main__ import parse_args
+
+    cfgf = tmp_path / "config.yaml"
+    cfgf.write_text(
+        "addr: 127.0.0.1:9999\n"
+        "runtime: proc\n"
+        "portRange: 50000-50100\n"      # camelCase accepted
+        "mock_gpus: 4\n"                # snake_case accepted
+        "xgmiProbe: true\n"
+        "apikey: sekrit\n"
+    )
+    c = parse_args(["--config", str(cfgf)])
+    assert c.addr == "127.0.0.1:9999"
+    assert c.runtime == "proc"
+    assert c.port_range == "50000-50100"
+    assert c.mock_gpus == 4
+    assert c.run_xgmi_probe is True
+    assert c.apikey == "sekrit"
+    # explicit flag beats the file
+    c = parse_args(["--config", str(cfgf), "--runtime", "mock", "--addr", "0.0.0.0:1"])
+    assert c.runtime == "mock" and c.addr == "0.0.0.0:1"
+    assert c.port_range == "50000-50100"  # file still layers under
+    # unknown keys are an error, not a silent ignore
+    bad = tmp_path / "bad.yaml"
+    bad.write_text("runtme: proc\n")
+    import pytest as _pytest
+
+    with _pytest.raises(SystemExit):
+        parse_args(["--config", str(bad)])
