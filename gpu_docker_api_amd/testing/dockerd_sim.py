@@ -31,7 +31,22 @@ from ..xerrors import ContainerExisted, ContainerNotExist, VolumeExisted
 
 class DockerdSim:
     def __init__(self, data_dir: str) -> None:
-        self.rt = ProcRuntime(base_dir=data_dir, use_cgroups=False)
+        # resolve GPU UUIDs -> indices for ROCR_VISIBLE_DEVICES like the
+        # daemon's proc runtime does; absent GPUs (CPU test boxes) resolve
+        # to None and the raw uuid passes through harmlessly
+        resolver = None
+        try:
+            from ..parallel.inventory import AmdSmiInventory
+
+            inv = AmdSmiInventory()
+            gpus = {g.uuid: g for g in inv.enumerate()}
+            if gpus:
+                resolver = lambda u: gpus.get(u)  # noqa: E731
+        except Exception:
+            resolver = None
+        self.rt = ProcRuntime(
+            base_dir=data_dir, use_cgroups=False, gpu_resolver=resolver or (lambda _u: None)
+        )
         # raw HostConfig per container, echoed back by inspect so clients
         # can verify device entries materialize
         self.host_configs: Dict[str, Dict[str, Any]] = {}
