@@ -36,6 +36,10 @@ class DockerdSim:
         # to None and the raw uuid passes through harmlessly
         resolver = None
         try:
+            import os as _os
+
+            if not _os.path.exists("/dev/kfd"):
+                raise RuntimeError("no GPU on this host")
             from ..parallel.inventory import AmdSmiInventory
 
             inv = AmdSmiInventory()
