@@ -527,3 +527,67 @@ def test_concurrent_clients_over_socket(dockerd_sim, tmp_path, run):
         await asyncio.gather(*(one(i) for i in range(6)))
 
     run(main())
+
+
+def test_service_chaos_over_engine_socket(dockerd_sim, tmp_path, run):
+    """A shorter run of the chaos model with the DOCKER runtime over the
+    live engine socket: the same invariants must hold when every container
+    operation is an Engine-API round-trip."""
+    import random
+
+    from gpu_docker_api_amd.models import ContainerRun, GpuPatch, MemoryPatch, PatchRequest
+    from gpu_docker_api_amd.routers.app import Daemon
+    from gpu_docker_api_amd.state.keys import Resource
+    from gpu_docker_api_amd.models.etcd import ContainerSpec
+    from gpu_docker_api_amd.xerrors import GdaError
+
+    cfg = make_config(tmp_path, runtime="docker", docker_socket=dockerd_sim,
+                      copy_engine="python")
+
+    async def main():
+        d = Daemon(cfg)
+        await d.start()
+        rng = random.Random(7)
+        tenants = ["wa", "wb", "wc"]
+
+        async def check():
+            want_gpu = 0
+            for name in tenants:
+                v = d.container_versions.get(name)
+                if v is None or f"{name}-{v}" in d.replicaset._released:
+                    continue
+                kv = await d.store.get_or_none(Resource.CONTAINERS, name)
+                spec = ContainerSpec.deserialize(kv.value)
+                assert spec.version == v
+                want_gpu += len(spec.gpu_uuids)
+            assert sum(d.gpu.get_gpu_status().values()) == want_gpu
+
+        for _ in range(45):
+            name = rng.choice(tenants)
+            op = rng.choice(["run", "patch_gpu", "patch_mem", "stop", "continue", "delete"])
+            try:
+                if op == "run":
+                    await d.replicaset.run_gpu_container(
+                        ContainerRun(image_name="img", replica_set_name=name,
+                                     gpu_count=rng.randint(0, 2), cmd=["sleep", "30"])
+                    )
+                elif op == "patch_gpu":
+                    await d.replicaset.patch_container(
+                        name, PatchRequest(gpu_patch=GpuPatch(gpu_count=rng.randint(0, 2)))
+                    )
+                elif op == "patch_mem":
+                    await d.replicaset.patch_container(
+                        name, PatchRequest(memory_patch=MemoryPatch(memory=f"{rng.randint(1, 3)}GB"))
+                    )
+                elif op == "stop":
+                    await d.replicaset.stop_container(name)
+                elif op == "continue":
+                    await d.replicaset.startup_container(name)
+                else:
+                    await d.replicaset.delete_container(name)
+            except GdaError:
+                pass
+            await check()
+        await d.stop()
+
+    run(main())
