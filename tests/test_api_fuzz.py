@@ -45,6 +45,15 @@ ROUTES = [
     ("PATCH", "/api/v1/volumes/fz/size"),
     ("POST", "/api/v1/resources/gpus/validate"),
     ("POST", "/api/v1/admin/compact"),
+    ("PUT", "/api/v1/images/fz"),
+]
+
+# GET routes with query params fuzzed separately (no body)
+GET_ROUTES = [
+    "/api/v1/replicaSet/fz/logs",
+    "/api/v1/replicaSet/fz/stats",
+    "/api/v1/replicaSet/fz/history",
+    "/api/v1/images",
 ]
 
 
@@ -57,6 +66,21 @@ ROUTES = [
 def test_any_body_yields_envelope(client, route, body):
     method, path = route
     r = client.request(method, path, json=body)
+    assert r.status_code == 200
+    payload = r.json()
+    assert set(payload) >= {"code", "msg", "data"}
+    assert payload["code"] in KNOWN_CODES
+
+
+@settings(
+    max_examples=60,
+    deadline=None,
+    suppress_health_check=[HealthCheck.function_scoped_fixture],
+)
+@given(path=st.sampled_from(GET_ROUTES), qk=st.text(max_size=8), qv=st.text(max_size=12))
+def test_get_routes_with_junk_query_yield_envelope(client, path, qk, qv):
+    params = {qk: qv, "tail": qv} if qk else {"tail": qv}
+    r = client.get(path, params=params)
     assert r.status_code == 200
     payload = r.json()
     assert set(payload) >= {"code", "msg", "data"}
