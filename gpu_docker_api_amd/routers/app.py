@@ -307,7 +307,9 @@ def _mount(app: FastAPI, d: Daemon) -> None:
             body = await request.json()
         except Exception:
             body = {}
-        path = (body or {}).get("path", "")
+        if not isinstance(body, dict):  # fuzz-found: a bare JSON scalar crashed
+            return error(Code.INVALID_PARAMS, detail="body must be a JSON object")
+        path = body.get("path", "")
         if not path:
             return error(Code.INVALID_PARAMS, detail="body needs {path: <local dir>}")
         try:
