@@ -193,19 +193,23 @@ class ProcRuntime(RuntimeDriver):
                 self.volumes[name] = VolumeState(name=name, mountpoint=mp, options=opts)
 
     @staticmethod
-    def _proc_starttime(pid: int) -> Optional[int]:
-        """starttime (clock ticks since boot) of a pid — stable identity
-        that survives pid reuse. Zombies (state Z — dead but unreaped by
-        their original parent, which may have been the previous daemon's
-        popen) count as gone."""
+    def _proc_stat(pid: int):
+        """(state_char, starttime) of a pid, or (None, None). starttime
+        (clock ticks since boot) is the stable identity that survives pid
+        reuse; zombies (Z — dead but unreaped by their original parent,
+        which may have been the previous daemon's popen) count as gone."""
         try:
             with open(f"/proc/{pid}/stat") as f:
                 fields = f.read().rsplit(") ", 1)[1].split()
                 if fields[0] in ("Z", "X"):
-                    return None
-                return int(fields[19])
+                    return None, None
+                return fields[0], int(fields[19])
         except (OSError, IndexError, ValueError):
-            return None
+            return None, None
+
+    @classmethod
+    def _proc_starttime(cls, pid: int) -> Optional[int]:
+        return cls._proc_stat(pid)[1]
 
     def _alive(self, p: _Proc) -> bool:
         if p.popen is not None:
@@ -295,7 +299,13 @@ class ProcRuntime(RuntimeDriver):
                     # later death gets normal (not stale) backoff
                     p.last_start_at = time.monotonic()
                     st.pid = pid
-                    st.running, st.status = True, "running"
+                    st.running = True
+                    # a SIGSTOPped group adopts as paused, not running
+                    state_char = self._proc_stat(pid)[0]
+                    if state_char in ("T", "t"):
+                        st.paused, st.status = True, "paused"
+                    else:
+                        st.status = "running"
                 else:
                     # it died while unsupervised; the restart policy decides
                     # whether the supervisor resurrects it (docker restarts

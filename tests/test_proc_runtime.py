@@ -413,3 +413,32 @@ def test_crash_loop_backoff(tmp_path, run):
         await rt.close()
 
     run(main())
+
+
+def test_adopted_paused_container_shows_paused(tmp_path, run):
+    """A container paused (SIGSTOP) before the daemon died must adopt as
+    paused, and unpause must resume it."""
+    from gpu_docker_api_amd.models.etcd import ContainerSpec
+    from gpu_docker_api_amd.runtime.proc import ProcRuntime
+
+    async def main():
+        rt1 = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        spec = ContainerSpec()
+        spec.config = {"Cmd": ["sleep", "60"]}
+        spec.container_name = "pz-1"
+        await rt1.create(spec)
+        await rt1.start("pz-1")
+        await rt1.pause("pz-1")
+        if rt1._supervisor is not None:
+            rt1._supervisor.cancel()
+
+        rt2 = ProcRuntime(base_dir=str(tmp_path), use_cgroups=False)
+        st = await rt2.inspect("pz-1")
+        assert st is not None and st.paused and st.status == "paused"
+        await rt2.unpause("pz-1")
+        st = await rt2.inspect("pz-1")
+        assert not st.paused and st.running
+        await rt2.remove("pz-1", force=True)
+        await rt2.close()
+
+    run(main())
